@@ -1,0 +1,122 @@
+"""Training workflow: the `pio train` core.
+
+Parity with the reference train path:
+- CreateWorkflow.main (core/.../workflow/CreateWorkflow.scala:136-280):
+  load engine.json variant, reflect EngineFactory, insert EngineInstance
+- CoreWorkflow.runTrain (core/.../workflow/CoreWorkflow.scala:45-102):
+  run engine.train, serialize models into the Models store, flip
+  EngineInstance INIT → COMPLETED
+- CleanupFunctions registry (workflow/CleanupFunctions.scala:26-63)
+
+The reference spark-submits a driver JVM; here training runs in-process (or
+is torchrun-launched for multi-GPU — see predictionio_amd.parallel).
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import traceback
+import uuid
+from typing import Any, Callable, Dict, List, Optional
+
+from predictionio_amd.controller.engine import Engine, EngineParams, get_engine
+from predictionio_amd.data import storage
+from predictionio_amd.data.events import utcnow
+from predictionio_amd.data.storage.base import EngineInstance, Model
+
+log = logging.getLogger(__name__)
+
+_cleanup_functions: List[Callable[[], None]] = []
+
+
+def add_cleanup(fn: Callable[[], None]) -> None:
+    """Register a cleanup hook run in finally of every workflow main
+    (CleanupFunctions.add)."""
+    _cleanup_functions.append(fn)
+
+
+def run_cleanup() -> None:
+    global _cleanup_functions
+    for fn in _cleanup_functions:
+        try:
+            fn()
+        except Exception:
+            log.exception("cleanup function failed")
+    _cleanup_functions = []
+
+
+def run_train(engine: Engine,
+              engine_params: EngineParams,
+              engine_id: str = "default",
+              engine_version: str = "0",
+              engine_variant: str = "default",
+              engine_factory: str = "",
+              batch: str = "",
+              env: Optional[Dict[str, str]] = None,
+              skip_sanity_check: bool = False,
+              verbose: bool = False) -> str:
+    """Train and persist; returns the engine-instance id
+    (CoreWorkflow.runTrain, CoreWorkflow.scala:45-102)."""
+    instances = storage.get_meta_data_engine_instances()
+    instance = EngineInstance(
+        id="", status="INIT", start_time=utcnow(), end_time=utcnow(),
+        engine_id=engine_id, engine_version=engine_version,
+        engine_variant=engine_variant, engine_factory=engine_factory,
+        batch=batch, env=env or {},
+        data_source_params=json.dumps(
+            {"name": engine_params.data_source_name,
+             "params": engine_params.data_source_params}),
+        preparator_params=json.dumps(
+            {"name": engine_params.preparator_name,
+             "params": engine_params.preparator_params}),
+        algorithms_params=json.dumps(
+            [{"name": n, "params": p}
+             for n, p in engine_params.algorithms_params]),
+        serving_params=json.dumps(
+            {"name": engine_params.serving_name,
+             "params": engine_params.serving_params}),
+    )
+    instance_id = instances.insert(instance)
+    try:
+        models = engine.train(engine_params,
+                              skip_sanity_check=skip_sanity_check)
+        blob = engine.make_serializable_models(
+            engine_params, instance_id, models)
+        storage.get_model_data_models().insert(Model(instance_id, blob))
+        instance.id = instance_id
+        instance.status = "COMPLETED"
+        instance.end_time = utcnow()
+        instances.update(instance)
+        log.info("Training completed: engine instance %s", instance_id)
+        return instance_id
+    except Exception:
+        instance.id = instance_id
+        instance.status = "FAILED"
+        instance.end_time = utcnow()
+        instances.update(instance)
+        raise
+    finally:
+        run_cleanup()
+
+
+def run_train_from_variant(variant: Dict[str, Any],
+                           engine_version: str = "0",
+                           batch: str = "",
+                           skip_sanity_check: bool = False) -> str:
+    """Entry point matching `pio train` with an engine.json variant dict:
+    {"id": ..., "engineFactory": "pkg.mod.Factory", "datasource": {...},
+     "algorithms": [...], ...} (CreateWorkflow.main, :136-280)."""
+    factory = variant.get("engineFactory")
+    if not factory:
+        raise ValueError("engine.json variant missing 'engineFactory'")
+    engine = get_engine(factory)
+    engine_params = engine.json_to_engine_params(variant)
+    return run_train(
+        engine, engine_params,
+        engine_id=variant.get("id", "default"),
+        engine_version=engine_version,
+        engine_variant=variant.get("id", "default"),
+        engine_factory=factory,
+        batch=batch,
+        skip_sanity_check=skip_sanity_check)
