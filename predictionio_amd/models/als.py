@@ -1,0 +1,154 @@
+"""ALS matrix factorization trainer — the framework's flagship compute path.
+
+Replaces Spark MLlib ALS (reference call sites: recommendation
+ALSAlgorithm.scala:75-86 explicit `ALS.train`, similarproduct
+ALSAlgorithm.scala:130-136 / ecommerce ECommAlgorithm.scala:123-133 implicit
+`ALS.trainImplicit`). Same semantics — ALS-WR lambda scaling for explicit,
+Hu-Koren confidence weighting for implicit, rank/iterations/lambda/alpha/seed
+hyperparameters — on an MI355X-native substrate:
+
+- ratings live as device-resident CSR shards (user-major + item-major)
+- each half-iteration is ONE fused HIP kernel launch (Gramian + Cholesky
+  per row, ops/csrc/als_kernels.hip) over the local row block
+- multi-GPU: user/item row-blocks per GPU; the fixed factor side is
+  all-gathered over RCCL/xGMI each half-iteration (the reference's MLlib
+  factor-block shuffle); no other communication.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional, Tuple
+
+import torch
+
+from predictionio_amd.ops import als as als_ops
+from predictionio_amd.parallel import dist as pdist
+
+
+@dataclass
+class ALSParams:
+    rank: int = 10
+    iterations: int = 10
+    lambda_: float = 0.01
+    alpha: float = 1.0          # implicit confidence weight
+    implicit: bool = False
+    seed: Optional[int] = None
+
+
+class ALSTrainer:
+    """Single-process trainer over one row-sharded data slice.
+
+    Distributed mode activates automatically when torch.distributed is
+    initialized: this rank owns users [u_lo, u_hi) and items [i_lo, i_hi);
+    `user_csr` indexes *global* item columns and `item_csr` global users.
+    """
+
+    def __init__(self, params: ALSParams, n_users: int, n_items: int,
+                 device: Optional[torch.device] = None):
+        self.p = params
+        self.n_users = n_users
+        self.n_items = n_items
+        self.device = device or (
+            torch.device("cuda") if torch.cuda.is_available()
+            else torch.device("cpu"))
+        world, rank = pdist.get_world_size(), pdist.get_rank()
+        self.u_lo, self.u_hi = pdist.block_bounds(n_users, world, rank)
+        self.i_lo, self.i_hi = pdist.block_bounds(n_items, world, rank)
+        self.user_csr = None   # (indptr, indices, values) local users x items
+        self.item_csr = None   # (indptr, indices, values) local items x users
+        self.X: Optional[torch.Tensor] = None  # local user factors
+        self.Y: Optional[torch.Tensor] = None  # local item factors
+
+    # ------------------------------------------------------------ data
+
+    def set_ratings(self, users: torch.Tensor, items: torch.Tensor,
+                    ratings: torch.Tensor) -> None:
+        """Load rating triples with *global* ids. In distributed mode each
+        rank must be fed (a) triples of its user block and (b) triples of
+        its item block (the setup-time all-to-all is the caller's job —
+        see engines; synthetic benches generate shards directly)."""
+        d = self.device
+        users = users.to(d)
+        items = items.to(d)
+        ratings = ratings.to(d)
+        u_mask = (users >= self.u_lo) & (users < self.u_hi)
+        self.user_csr = als_ops.build_csr(
+            users[u_mask] - self.u_lo, items[u_mask], ratings[u_mask],
+            self.u_hi - self.u_lo)
+        i_mask = (items >= self.i_lo) & (items < self.i_hi)
+        self.item_csr = als_ops.build_csr(
+            items[i_mask] - self.i_lo, users[i_mask], ratings[i_mask],
+            self.i_hi - self.i_lo)
+
+    def set_ratings_sharded(self, user_shard, item_shard) -> None:
+        """Directly install pre-sharded triples: user_shard = (u, i, r) for
+        this rank's users (u local-indexed), item_shard = (i, u, r) for this
+        rank's items (i local-indexed)."""
+        u, i, r = user_shard
+        self.user_csr = als_ops.build_csr(
+            u.to(self.device), i.to(self.device), r.to(self.device),
+            self.u_hi - self.u_lo)
+        i2, u2, r2 = item_shard
+        self.item_csr = als_ops.build_csr(
+            i2.to(self.device), u2.to(self.device), r2.to(self.device),
+            self.i_hi - self.i_lo)
+
+    def init_factors(self) -> None:
+        """Random init, matching MLlib's scheme (unit-scaled gaussian /
+        sqrt(rank) keeps initial Gramians well-conditioned)."""
+        g = torch.Generator(device="cpu")
+        if self.p.seed is not None:
+            g.manual_seed(self.p.seed + pdist.get_rank())
+        f = self.p.rank
+        self.X = (torch.randn((self.u_hi - self.u_lo, f), generator=g)
+                  / (f ** 0.5)).to(self.device)
+        self.Y = (torch.randn((self.i_hi - self.i_lo, f), generator=g)
+                  / (f ** 0.5)).to(self.device)
+
+    # ------------------------------------------------------------ training
+
+    def _half_step(self, csr, fixed_local: torch.Tensor,
+                   n_fixed: int) -> torch.Tensor:
+        """One half-iteration: all-gather the fixed side, fused solve."""
+        fixed_full = pdist.all_gather_rows(fixed_local, n_fixed)
+        yty = als_ops.gramian(fixed_full) if self.p.implicit else None
+        indptr, indices, values = csr
+        return als_ops.als_solve(
+            indptr, indices, values, fixed_full, YtY=yty,
+            lam=self.p.lambda_, alpha=self.p.alpha,
+            implicit=self.p.implicit, wr_scale=not self.p.implicit)
+
+    def step(self) -> None:
+        """One full ALS iteration (user half-step then item half-step)."""
+        self.X = self._half_step(self.user_csr, self.Y, self.n_items)
+        self.Y = self._half_step(self.item_csr, self.X, self.n_users)
+
+    def fit(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        if self.X is None:
+            self.init_factors()
+        for _ in range(self.p.iterations):
+            self.step()
+        return self.X, self.Y
+
+    # ------------------------------------------------------------ model out
+
+    def gather_factors(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Full (X, Y) on every rank (checkpoint/serving path)."""
+        X = pdist.all_gather_rows(self.X, self.n_users)
+        Y = pdist.all_gather_rows(self.Y, self.n_items)
+        return X, Y
+
+    def local_nnz(self) -> int:
+        return int(self.user_csr[1].numel() + self.item_csr[1].numel())
+
+
+def train_als(users: torch.Tensor, items: torch.Tensor,
+              ratings: torch.Tensor, n_users: int, n_items: int,
+              params: ALSParams,
+              device: Optional[torch.device] = None
+              ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Convenience single-call trainer (local factors returned)."""
+    t = ALSTrainer(params, n_users, n_items, device)
+    t.set_ratings(users, items, ratings)
+    return t.fit()

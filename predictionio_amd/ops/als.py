@@ -1,0 +1,130 @@
+"""ALS half-iteration solve: HIP kernel on GPU, torch reference on CPU.
+
+Replaces MLlib's normal-equation construction + CholeskySolver (reference:
+ALS.train/trainImplicit call sites, SURVEY.md §2.7/2.9 K1+K2). The CPU path
+is the numerics reference the GPU kernel is validated against.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+SUPPORTED_RANKS = (16, 32, 64, 128)
+
+
+def pad_rank(f: int) -> int:
+    for s in SUPPORTED_RANKS:
+        if f <= s:
+            return s
+    raise ValueError(f"rank {f} > 128 not supported")
+
+
+def gramian(Y: torch.Tensor) -> torch.Tensor:
+    """YtY (f x f) — a plain library GEMM (hipBLASLt on device)."""
+    return Y.t() @ Y
+
+
+def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
+              values: torch.Tensor, Y: torch.Tensor,
+              YtY: Optional[torch.Tensor] = None,
+              lam: float = 0.01, alpha: float = 1.0,
+              implicit: bool = False, wr_scale: bool = True) -> torch.Tensor:
+    """Solve all rows of one ALS half-iteration.
+
+    explicit: (sum y y^T + lam*nnz*I) x = sum r*y      (ALS-WR, like MLlib)
+    implicit: (YtY + sum alpha*r y y^T + lam*I) x = sum (1+alpha*r) y
+              (Hu-Koren; YtY required)
+    """
+    if implicit and YtY is None:
+        YtY = gramian(Y)
+    if Y.is_cuda:
+        from predictionio_amd.ops import hip_ext
+        f = Y.shape[1]
+        pf = pad_rank(f)
+        Yp = Y if pf == f else torch.nn.functional.pad(Y, (0, pf - f))
+        YtYp = None
+        if YtY is not None:
+            YtYp = (YtY if pf == f
+                    else torch.nn.functional.pad(YtY, (0, pf - f, 0, pf - f)))
+        X = hip_ext().als_solve(
+            indptr.contiguous(), indices.contiguous(), values.contiguous(),
+            Yp.contiguous(),
+            YtYp.contiguous() if YtYp is not None else None,
+            float(lam), float(alpha), bool(implicit), bool(wr_scale))
+        return X[:, :f].contiguous() if pf != f else X
+    return als_solve_ref(indptr, indices, values, Y, YtY, lam, alpha,
+                         implicit, wr_scale)
+
+
+def als_solve_ref(indptr, indices, values, Y, YtY=None, lam=0.01, alpha=1.0,
+                  implicit=False, wr_scale=True) -> torch.Tensor:
+    """Pure-torch fp32 reference (used on CPU and in GPU numerics tests)."""
+    n_rows = indptr.shape[0] - 1
+    f = Y.shape[1]
+    X = torch.zeros((n_rows, f), dtype=Y.dtype, device=Y.device)
+    eye = torch.eye(f, dtype=Y.dtype, device=Y.device)
+    ip = indptr.tolist()
+    for r in range(n_rows):
+        s, e = ip[r], ip[r + 1]
+        nnz = e - s
+        if implicit:
+            A = (YtY if YtY is not None else gramian(Y)).clone()
+            b = torch.zeros(f, dtype=Y.dtype, device=Y.device)
+            if nnz:
+                cols = indices[s:e].long()
+                Yr = Y[cols]
+                v = values[s:e]
+                A = A + (Yr.t() * (alpha * v)) @ Yr
+                b = Yr.t() @ (1.0 + alpha * v)
+            A = A + lam * eye
+        else:
+            if nnz == 0:
+                continue
+            cols = indices[s:e].long()
+            Yr = Y[cols]
+            v = values[s:e]
+            A = Yr.t() @ Yr + lam * (nnz if wr_scale else 1.0) * eye
+            b = Yr.t() @ v
+        X[r] = torch.linalg.solve(A, b)
+    return X
+
+
+def build_csr(rows: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
+              n_rows: int) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """(row, col, val) triples → CSR (indptr i64, indices i32, values f32).
+
+    Device-side sort-based build — torch.sort on GPU.
+    """
+    order = torch.argsort(rows)
+    r = rows[order]
+    indices = cols[order].to(torch.int32)
+    values = vals[order].to(torch.float32)
+    counts = torch.bincount(r.long(), minlength=n_rows)
+    indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=rows.device)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    return indptr, indices, values
+
+
+def aggregate_ratings(rows: torch.Tensor, cols: torch.Tensor,
+                      vals: torch.Tensor, n_cols: int,
+                      mode: str = "sum"):
+    """Dedup (row, col) pairs: 'sum' (e-commerce reduceByKey semantics,
+    ECommAlgorithm.scala:168-205) or 'latest' (similarproduct latest-wins,
+    ALSAlgorithm.scala:88-120 — caller passes vals in time order)."""
+    key = rows.long() * n_cols + cols.long()
+    if mode == "sum":
+        uk, inv = torch.unique(key, return_inverse=True)
+        out = torch.zeros(uk.numel(), dtype=torch.float32, device=vals.device)
+        out.scatter_add_(0, inv, vals.float())
+    elif mode == "latest":
+        # stable sort keeps time order within key; take last occurrence
+        sk, order = torch.sort(key, stable=True)
+        last = torch.ones(sk.numel(), dtype=torch.bool, device=key.device)
+        last[:-1] = sk[1:] != sk[:-1]
+        uk = sk[last]
+        out = vals[order][last].float()
+    else:
+        raise ValueError(mode)
+    return (uk // n_cols).to(rows.dtype), (uk % n_cols).to(cols.dtype), out

@@ -1,0 +1,197 @@
+// Fused masked top-K scoring for MI355X (gfx950, CDNA4).
+//
+// Replaces the reference's serve-time scoring loops:
+//  - recommendation `recommendProductsWithFilter`: blas.ddot per item +
+//    scored.top(num) (examples/.../ALSModel.scala:44-60)
+//  - similarproduct cosine + PriorityQueue top-N
+//    (examples/.../ALSAlgorithm.scala:168-242)
+//  - ecommerce predictKnownUser / predictSimilar
+//    (examples/.../ECommAlgorithm.scala:471-506, 541-599)
+//
+// Design (MI355X-first): scores = Xq . Y^T fused with masking and an
+// on-device per-slice top-K reduction, never materializing the B x N score
+// matrix (N up to 10^7). Grid = (item_slices, user_blocks); each 256-thread
+// workgroup stages a 64-item chunk of Y in LDS (coalesced, +1-padded
+// against bank conflicts), computes dots for UPB users per wave, and
+// maintains a per-user running top-K list in LDS with a wave-ballot
+// insertion filter. Phase 2 (merging the per-slice candidates) is a small
+// torch.topk on [B, slices*K].
+//
+// Masks:
+//  - item_mask: optional uint8[N], 1 = globally banned (e.g. unavailable
+//    items, category filter precomputed on device)
+//  - per-user banned list (seen/blacklisted items): CSR int32, binary
+//    search per candidate insertion (lists are short; L1-resident)
+
+#include <float.h>
+#include <hip/hip_runtime.h>
+
+#define TK_CHUNK 64   // items staged per LDS pass (= wave width)
+#define TK_MAXK 64    // max supported K
+
+__device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
+  int lo = 0, hi = n - 1;
+  while (lo <= hi) {
+    int mid = (lo + hi) >> 1;
+    int v = arr[mid];
+    if (v == x) return true;
+    if (v < x) lo = mid + 1; else hi = mid - 1;
+  }
+  return false;
+}
+
+template <int F, int UPW>  // UPW = users per wave
+__global__ __launch_bounds__(256) void topk_score_kernel(
+    const float* __restrict__ Xq,        // B x F query vectors
+    const float* __restrict__ Y,         // N x F item factors
+    const uint8_t* __restrict__ item_mask,       // N or nullptr
+    const long long* __restrict__ ban_indptr,    // B+1 or nullptr
+    const int* __restrict__ ban_indices,         // sorted per user
+    float* __restrict__ out_val,         // B x n_slices x K
+    int* __restrict__ out_idx,           // B x n_slices x K
+    int B, long long N, int K, int n_slices, int item_base)
+{
+  constexpr int WAVES = 4;
+  constexpr int UPB = WAVES * UPW;  // users per block
+
+  __shared__ float ys[TK_CHUNK][F + 1];
+  __shared__ float xs[UPB][F + 1];
+  __shared__ float topv[UPB][TK_MAXK];
+  __shared__ int topi[UPB][TK_MAXK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+
+  const int slice = blockIdx.x;
+  const int ublock = blockIdx.y;
+  const long long u0 = (long long)ublock * UPB;
+
+  // item range of this slice
+  const long long per = (N + n_slices - 1) / n_slices;
+  const long long it0 = (long long)slice * per;
+  const long long it1 = min(N, it0 + per);
+
+  // stage user query vectors + init top-K state
+  for (int e = tid; e < UPB * F; e += 256) {
+    const int u = e / F;
+    const int k = e % F;
+    xs[u][k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
+  }
+  for (int e = tid; e < UPB * K; e += 256) {
+    topv[e / K][e % K] = -FLT_MAX;
+    topi[e / K][e % K] = -1;
+  }
+  __syncthreads();
+
+  // running K-th-best threshold per user, register-replicated across the
+  // wave (intra-wave LDS cross-lane communication is not ordered without a
+  // barrier — shuffles are)
+  float uth[UPW];
+#pragma unroll
+  for (int uu = 0; uu < UPW; ++uu) uth[uu] = -FLT_MAX;
+
+  for (long long base = it0; base < it1; base += TK_CHUNK) {
+    const int cn = (int)min((long long)TK_CHUNK, it1 - base);
+    __syncthreads();
+    // coalesced stage of cn item rows
+    for (int e = tid; e < cn * F; e += 256) {
+      const int c = e / F;
+      const int k = e % F;
+      ys[c][k] = Y[(base + c) * F + k];
+    }
+    __syncthreads();
+
+    // each wave handles its UPW users; lane l scores item l of the chunk
+#pragma unroll
+    for (int uu = 0; uu < UPW; ++uu) {
+      const int u = wave * UPW + uu;
+      const long long guser = u0 + u;
+      float s = -FLT_MAX;
+      const long long item = base + lane;
+      if (lane < cn && guser < B) {
+        bool banned = false;
+        if (item_mask != nullptr && item_mask[item]) banned = true;
+        if (!banned) {
+          float acc = 0.f;
+#pragma unroll 8
+          for (int k = 0; k < F; ++k) acc = fmaf(xs[u][k], ys[lane][k], acc);
+          s = acc;
+        }
+      }
+      // wave-ballot insertion: only lanes beating the running threshold
+      unsigned long long mask = __ballot(s > uth[uu]);
+      while (mask) {
+        const int src = __ffsll(mask) - 1;
+        mask &= mask - 1;
+        const float v = __shfl(s, src);
+        const long long cand = base + src;
+        float nth = uth[uu];
+        if (lane == 0) {
+          // optional per-user banned-list check (short sorted list);
+          // topv/topi rows of this user are touched by lane 0 only inside
+          // the scan loop, so no cross-lane LDS hazard here
+          bool banned = false;
+          if (ban_indptr != nullptr) {
+            const long long bs0 = ban_indptr[guser];
+            const int bn = (int)(ban_indptr[guser + 1] - bs0);
+            banned = in_sorted(ban_indices + bs0, bn, (int)(cand + item_base));
+          }
+          if (!banned && v > nth) {
+            // replace current min of the K-list
+            int mi = 0;
+            float mv = topv[u][0];
+            for (int q = 1; q < K; ++q)
+              if (topv[u][q] < mv) { mv = topv[u][q]; mi = q; }
+            topv[u][mi] = v;
+            topi[u][mi] = (int)(cand + item_base);
+            // new threshold = K-th best = new min of the list
+            float nm = topv[u][0];
+            for (int q = 1; q < K; ++q) nm = fminf(nm, topv[u][q]);
+            nth = nm;
+          }
+        }
+        uth[uu] = __shfl(nth, 0);
+        if (mask) mask &= __ballot(s > uth[uu]);
+      }
+    }
+  }
+  __syncthreads();
+
+  // write out per-slice candidates
+  for (int e = tid; e < UPB * K; e += 256) {
+    const int u = e / K;
+    const int q = e % K;
+    const long long guser = u0 + u;
+    if (guser < B) {
+      const long long o = (guser * n_slices + slice) * K + q;
+      out_val[o] = topv[u][q];
+      out_idx[o] = topi[u][q];
+    }
+  }
+}
+
+extern "C" void launch_topk_score(
+    const float* Xq, const float* Y, const uint8_t* item_mask,
+    const long long* ban_indptr, const int* ban_indices,
+    float* out_val, int* out_idx,
+    int B, long long N, int f, int K, int n_slices, int item_base,
+    hipStream_t stream)
+{
+  constexpr int UPW = 2;
+  const int UPB = 4 * UPW;
+  dim3 grid(n_slices, (B + UPB - 1) / UPB);
+  dim3 block(256);
+#define LAUNCH(FF)                                                        \
+  hipLaunchKernelGGL((topk_score_kernel<FF, UPW>), grid, block, 0, stream,\
+                     Xq, Y, item_mask, ban_indptr, ban_indices, out_val,  \
+                     out_idx, B, N, K, n_slices, item_base)
+  switch (f) {
+    case 16: LAUNCH(16); break;
+    case 32: LAUNCH(32); break;
+    case 64: LAUNCH(64); break;
+    case 128: LAUNCH(128); break;
+    default: break;
+  }
+#undef LAUNCH
+}

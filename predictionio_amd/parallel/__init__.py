@@ -1,0 +1,17 @@
+"""Multi-GPU data parallelism over RCCL/xGMI.
+
+Replaces the reference's Spark-shuffle data plane: the
+ALS half-iteration factor-block shuffle becomes an RCCL all-gather of the
+fixed factor side over xGMI; driver `collect` disappears (factors stay
+resident in HBM); top-K serving merges per-GPU candidates with a K-wide
+gather. One process per GPU via torch.distributed (backend "nccl" IS RCCL
+on ROCm; tests use "gloo" on CPU).
+"""
+
+from predictionio_amd.parallel.dist import (
+    all_gather_rows, block_bounds, get_rank, get_world_size, init_from_env,
+    is_distributed,
+)
+
+__all__ = ["all_gather_rows", "block_bounds", "get_rank", "get_world_size",
+           "init_from_env", "is_distributed"]

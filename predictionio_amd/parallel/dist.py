@@ -1,0 +1,158 @@
+"""torch.distributed helpers: process-group init, row-block sharding,
+factor all-gather.
+
+MI355X topology notes: xGMI is point-to-point — 7 links
+x ~153 GB/s per GPU — so the per-half-iteration factor exchange is a direct
+all-gather of row-shards (each GPU sends its shard to 7 peers
+independently), not a ring all-reduce. `all_gather_into_tensor` over RCCL
+maps to exactly that.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+def is_distributed() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if is_distributed() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if is_distributed() else 1
+
+
+def init_from_env(backend: str | None = None,
+                  timeout_s: float = 600.0) -> Tuple[int, int]:
+    """Initialize the default process group from torchrun env vars
+    (RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT). Backend defaults to
+    nccl(=RCCL) when a GPU is visible, else gloo. Returns (rank, world)."""
+    if is_distributed():
+        return get_rank(), get_world_size()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return 0, 1
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    dist.init_process_group(
+        backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
+    if backend == "nccl":
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    return get_rank(), get_world_size()
+
+
+def block_bounds(n: int, world: int, rank: int) -> Tuple[int, int]:
+    """Contiguous row-block [lo, hi) of rank; first `n % world` blocks get
+    one extra row."""
+    base, rem = divmod(n, world)
+    lo = rank * base + min(rank, rem)
+    hi = lo + base + (1 if rank < rem else 0)
+    return lo, hi
+
+
+def all_gather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
+    """All-gather row-sharded [n_local, f] tensors into [n_total, f].
+
+    Shards may be uneven (block_bounds); pads to the max shard for
+    all_gather_into_tensor (single fused RCCL call over xGMI), then
+    reassembles. No-op when not distributed."""
+    if not is_distributed():
+        return local
+    world = get_world_size()
+    f = local.shape[1]
+    max_rows = (n_total + world - 1) // world
+    padded = local
+    if local.shape[0] < max_rows:
+        padded = torch.empty((max_rows, f), dtype=local.dtype,
+                             device=local.device)
+        padded[:local.shape[0]] = local
+        padded[local.shape[0]:] = 0
+    else:
+        padded = local.contiguous()
+    out = torch.empty((world * max_rows, f), dtype=local.dtype,
+                      device=local.device)
+    dist.all_gather_into_tensor(out, padded)
+    # reassemble uneven blocks
+    pieces: List[torch.Tensor] = []
+    for r in range(world):
+        lo, hi = block_bounds(n_total, world, r)
+        pieces.append(out[r * max_rows: r * max_rows + (hi - lo)])
+    return torch.cat(pieces, dim=0) if world > 1 else out[:n_total]
+
+
+def exchange_triples(rows: torch.Tensor, cols: torch.Tensor,
+                     vals: torch.Tensor, n_cols: int):
+    """Repartition (row, col, val) triples so each rank receives every
+    triple whose `col` falls in its col-block — the one-time setup shuffle
+    that replaces Spark's rating-block exchange. Uses
+    all_to_all_single over RCCL on GPU; falls back to all_gather on gloo
+    (gloo lacks all_to_all). Returns (rows, cols, vals) of the local block
+    with GLOBAL col ids."""
+    if not is_distributed():
+        return rows, cols, vals
+    world = get_world_size()
+    rank = get_rank()
+    device = rows.device
+    # destination rank per triple
+    boundary = torch.tensor(
+        [block_bounds(n_cols, world, r)[1] for r in range(world)],
+        device=device, dtype=torch.int64)
+    dest = torch.searchsorted(boundary, cols.long(), right=True)
+    order = torch.argsort(dest)
+    rows_s, cols_s, vals_s = rows[order], cols[order], vals[order]
+    counts = torch.bincount(dest, minlength=world)
+    if dist.get_backend() == "gloo":
+        gathered: list = [None] * world
+        dist.all_gather_object(
+            gathered, (rows_s.cpu(), cols_s.cpu(), vals_s.cpu(),
+                       counts.cpu()))
+        lo, hi = block_bounds(n_cols, world, rank)
+        out_r, out_c, out_v = [], [], []
+        for r_, (rr, cc, vv, cn) in enumerate(gathered):
+            off = int(cn[:rank].sum())
+            n = int(cn[rank])
+            out_r.append(rr[off:off + n])
+            out_c.append(cc[off:off + n])
+            out_v.append(vv[off:off + n])
+        return (torch.cat(out_r).to(device), torch.cat(out_c).to(device),
+                torch.cat(out_v).to(device))
+    # RCCL path: exchange counts, then all_to_all_single per array
+    in_splits = counts.tolist()
+    counts_all = torch.empty((world, world), dtype=torch.int64,
+                             device=device)
+    dist.all_gather_into_tensor(counts_all.view(-1),
+                                counts.to(device, torch.int64))
+    out_splits = counts_all[:, rank].tolist()
+    total_out = sum(out_splits)
+
+    def a2a(t: torch.Tensor) -> torch.Tensor:
+        out = torch.empty(total_out, dtype=t.dtype, device=device)
+        dist.all_to_all_single(out, t.contiguous(),
+                               output_split_sizes=out_splits,
+                               input_split_sizes=in_splits)
+        return out
+
+    return a2a(rows_s), a2a(cols_s), a2a(vals_s)
+
+
+def barrier() -> None:
+    if is_distributed():
+        dist.barrier()
+
+
+def max_scalar(x: float, device=None) -> float:
+    """MAX over ranks of a host scalar (for timing: take the slowest rank)."""
+    if not is_distributed():
+        return x
+    t = torch.tensor([x], dtype=torch.float64,
+                     device=device if device is not None else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())

@@ -1,0 +1,169 @@
+"""CPU tests of the compute-op reference implementations + ALS trainer."""
+
+import math
+
+import pytest
+import torch
+
+from predictionio_amd.models.als import ALSParams, ALSTrainer, train_als
+from predictionio_amd.ops import als as als_ops
+from predictionio_amd.ops import topk as topk_ops
+
+
+def synth(n_users=50, n_items=30, f=8, nnz_per_user=10, seed=0,
+          implicit=False):
+    g = torch.Generator().manual_seed(seed)
+    Xt = torch.randn((n_users, f), generator=g) / math.sqrt(f)
+    Yt = torch.randn((n_items, f), generator=g) / math.sqrt(f)
+    users = torch.arange(n_users).repeat_interleave(nnz_per_user)
+    items = torch.randint(0, n_items, (n_users * nnz_per_user,), generator=g)
+    # dedup (keep one rating per (u, i))
+    key = users * n_items + items
+    uniq = torch.unique(key)
+    users, items = (uniq // n_items).int(), (uniq % n_items).int()
+    r_true = (Xt[users.long()] * Yt[items.long()]).sum(1)
+    ratings = torch.ones_like(r_true) if implicit else r_true + 3.0
+    return users, items, ratings.float(), Xt, Yt
+
+
+class TestCSRBuild:
+    def test_build_csr(self):
+        rows = torch.tensor([2, 0, 2, 1], dtype=torch.int32)
+        cols = torch.tensor([5, 1, 3, 2], dtype=torch.int32)
+        vals = torch.tensor([1., 2., 3., 4.])
+        indptr, indices, values = als_ops.build_csr(rows, cols, vals, 4)
+        assert indptr.tolist() == [0, 1, 2, 4, 4]
+        assert indices[0].item() == 1 and values[0].item() == 2.0
+        assert sorted(indices[2:4].tolist()) == [3, 5]
+
+    def test_aggregate_sum(self):
+        rows = torch.tensor([0, 0, 1], dtype=torch.int32)
+        cols = torch.tensor([1, 1, 2], dtype=torch.int32)
+        vals = torch.tensor([1., 2., 5.])
+        r, c, v = als_ops.aggregate_ratings(rows, cols, vals, 10, "sum")
+        d = {(int(a), int(b)): float(x) for a, b, x in zip(r, c, v)}
+        assert d == {(0, 1): 3.0, (1, 2): 5.0}
+
+    def test_aggregate_latest(self):
+        rows = torch.tensor([0, 0, 1], dtype=torch.int32)
+        cols = torch.tensor([1, 1, 2], dtype=torch.int32)
+        vals = torch.tensor([1., 2., 5.])  # time order
+        r, c, v = als_ops.aggregate_ratings(rows, cols, vals, 10, "latest")
+        d = {(int(a), int(b)): float(x) for a, b, x in zip(r, c, v)}
+        assert d == {(0, 1): 2.0, (1, 2): 5.0}
+
+
+class TestALSSolveRef:
+    def test_explicit_solves_normal_equations(self):
+        users, items, ratings, _, _ = synth()
+        n_users, n_items, f = 50, 30, 8
+        Y = torch.randn((n_items, f)) / math.sqrt(f)
+        indptr, indices, values = als_ops.build_csr(users, items, ratings,
+                                                    n_users)
+        lam = 0.05
+        X = als_ops.als_solve_ref(indptr, indices, values, Y, lam=lam)
+        # check row 0 against a hand-built solve
+        s, e = indptr[0].item(), indptr[1].item()
+        Yr = Y[indices[s:e].long()]
+        A = Yr.t() @ Yr + lam * (e - s) * torch.eye(f)
+        b = Yr.t() @ values[s:e]
+        assert torch.allclose(X[0], torch.linalg.solve(A, b), atol=1e-5)
+
+    def test_implicit_solves_hu_koren(self):
+        users, items, ratings, _, _ = synth(implicit=True)
+        n_users, n_items, f = 50, 30, 8
+        Y = torch.randn((n_items, f)) / math.sqrt(f)
+        indptr, indices, values = als_ops.build_csr(users, items, ratings,
+                                                    n_users)
+        lam, alpha = 0.05, 2.0
+        YtY = als_ops.gramian(Y)
+        X = als_ops.als_solve_ref(indptr, indices, values, Y, YtY=YtY,
+                                  lam=lam, alpha=alpha, implicit=True)
+        s, e = indptr[3].item(), indptr[4].item()
+        Yr = Y[indices[s:e].long()]
+        v = values[s:e]
+        A = YtY + (Yr.t() * (alpha * v)) @ Yr + lam * torch.eye(f)
+        b = Yr.t() @ (1 + alpha * v)
+        assert torch.allclose(X[3], torch.linalg.solve(A, b), atol=1e-5)
+
+
+class TestALSTrainer:
+    def test_explicit_reduces_rmse(self):
+        users, items, ratings, _, _ = synth(n_users=80, n_items=40, f=6,
+                                            nnz_per_user=15)
+        p = ALSParams(rank=6, iterations=8, lambda_=0.01, seed=1)
+        X, Y = train_als(users, items, ratings, 80, 40, p,
+                         device=torch.device("cpu"))
+        pred = (X[users.long()] * Y[items.long()]).sum(1)
+        rmse = ((pred - ratings) ** 2).mean().sqrt().item()
+        assert rmse < 0.15, f"train RMSE too high: {rmse}"
+
+    def test_implicit_ranks_observed_higher(self):
+        users, items, ratings, _, _ = synth(n_users=60, n_items=30, f=6,
+                                            implicit=True, nnz_per_user=8)
+        p = ALSParams(rank=6, iterations=10, lambda_=0.05, alpha=5.0,
+                      implicit=True, seed=2)
+        X, Y = train_als(users, items, ratings, 60, 30, p,
+                         device=torch.device("cpu"))
+        scores = X @ Y.t()
+        obs = scores[users.long(), items.long()].mean()
+        assert obs > scores.mean() + 0.1
+
+    def test_step_equivalence_with_fit(self):
+        users, items, ratings, _, _ = synth()
+        p = ALSParams(rank=8, iterations=3, seed=3)
+        t1 = ALSTrainer(p, 50, 30, torch.device("cpu"))
+        t1.set_ratings(users, items, ratings)
+        X1, Y1 = t1.fit()
+        t2 = ALSTrainer(p, 50, 30, torch.device("cpu"))
+        t2.set_ratings(users, items, ratings)
+        t2.init_factors()
+        for _ in range(3):
+            t2.step()
+        assert torch.allclose(X1, t2.X) and torch.allclose(Y1, t2.Y)
+
+
+class TestTopKRef:
+    def test_basic_topk(self):
+        Xq = torch.tensor([[1.0, 0.0], [0.0, 1.0]])
+        Y = torch.tensor([[1., 0.], [2., 0.], [0., 3.], [0., 0.5]])
+        vals, idxs = topk_ops.topk_score_ref(Xq, Y, 2)
+        assert idxs[0].tolist() == [1, 0]
+        assert idxs[1].tolist() == [2, 3]
+
+    def test_global_mask(self):
+        Xq = torch.tensor([[1.0, 0.0]])
+        Y = torch.tensor([[1., 0.], [2., 0.], [3., 0.]])
+        mask = torch.tensor([0, 0, 1], dtype=torch.uint8)
+        vals, idxs = topk_ops.topk_score_ref(Xq, Y, 2, item_mask=mask)
+        assert idxs[0].tolist() == [1, 0]
+
+    def test_per_user_ban(self):
+        Xq = torch.eye(2)
+        Y = torch.tensor([[3., 3.], [2., 2.], [1., 1.]])
+        bi = torch.tensor([0, 1, 1], dtype=torch.int64)  # user0 bans item0
+        bx = torch.tensor([0], dtype=torch.int32)
+        vals, idxs = topk_ops.topk_score_ref(Xq, Y, 1, ban_indptr=bi,
+                                             ban_indices=bx)
+        assert idxs[0].item() == 1
+        assert idxs[1].item() == 0
+
+    def test_empty_slots(self):
+        Xq = torch.tensor([[1.0]])
+        Y = torch.tensor([[1.0], [2.0]])
+        mask = torch.tensor([1, 1], dtype=torch.uint8)
+        vals, idxs = topk_ops.topk_score_ref(Xq, Y, 2, item_mask=mask)
+        assert idxs[0].tolist() == [-1, -1]
+
+    def test_cosine_collapse(self):
+        g = torch.Generator().manual_seed(0)
+        Y = torch.randn((20, 8), generator=g)
+        Yn = topk_ops.normalize_rows(Y)
+        qitems = [3, 7]
+        qvec = Yn[qitems].sum(0)
+        vals, idxs = topk_ops.cosine_topk(qvec, Yn, 5)
+        # brute-force cosine sum
+        import torch.nn.functional as F
+        ref = sum(F.cosine_similarity(Y[q].unsqueeze(0), Y) for q in qitems)
+        rv, ri = torch.topk(ref, 5)
+        assert idxs[0].tolist() == ri.tolist()
