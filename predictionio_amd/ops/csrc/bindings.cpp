@@ -6,8 +6,8 @@
 #include <torch/extension.h>
 
 #include <ATen/hip/HIPContext.h>
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
 extern "C" void launch_als_solve(
@@ -61,8 +61,8 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
     yty_ptr = YtY->data_ptr<float>();
   }
   auto X = torch::empty({n_rows, f}, Y.options());
-  c10::hip::HIPGuard guard(Y.device());
-  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
+  hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   launch_als_solve(reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()), indices.data_ptr<int>(),
                    values.data_ptr<float>(), Y.data_ptr<float>(), yty_ptr,
                    X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
@@ -111,8 +111,8 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
   auto out_val = torch::empty({B, n_slices * K}, Xq.options());
   auto out_idx = torch::empty({B, n_slices * K},
                               Xq.options().dtype(torch::kInt32));
-  c10::hip::HIPGuard guard(Y.device());
-  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
+  hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   launch_topk_score(Xq.data_ptr<float>(), Y.data_ptr<float>(), mask_ptr,
                     bi_ptr, bx_ptr, out_val.data_ptr<float>(),
                     out_idx.data_ptr<int>(), (int)B, (long long)N, (int)f,
