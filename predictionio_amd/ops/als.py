@@ -37,8 +37,10 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
     implicit: (YtY + sum alpha*r y y^T + lam*I) x = sum (1+alpha*r) y
               (Hu-Koren; YtY required)
     """
-    if implicit and YtY is None:
-        YtY = gramian(Y)
+    if implicit:
+        wr_scale = False  # Hu-Koren regularizes with plain lambda*I
+        if YtY is None:
+            YtY = gramian(Y)
     if Y.is_cuda:
         from predictionio_amd.ops import hip_ext
         f = Y.shape[1]
