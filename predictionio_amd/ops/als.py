@@ -26,16 +26,33 @@ def gramian(Y: torch.Tensor) -> torch.Tensor:
     return Y.t() @ Y
 
 
+def woodbury_w(Y: torch.Tensor, YtY: torch.Tensor,
+               lam: float) -> torch.Tensor:
+    """W = (YtY + lam I)^-1 Y — the per-half-iteration precompute that
+    enables the per-row Woodbury fast path (als_woodbury_kernel). One
+    64x64 Cholesky + one triangular-solve GEMM over all items (rocBLAS)."""
+    f = Y.shape[1]
+    B = YtY + lam * torch.eye(f, dtype=Y.dtype, device=Y.device)
+    L = torch.linalg.cholesky(B)
+    return torch.cholesky_solve(Y.t().contiguous(), L).t().contiguous()
+
+
 def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
               values: torch.Tensor, Y: torch.Tensor,
               YtY: Optional[torch.Tensor] = None,
               lam: float = 0.01, alpha: float = 1.0,
-              implicit: bool = False, wr_scale: bool = True) -> torch.Tensor:
+              implicit: bool = False, wr_scale: bool = True,
+              W: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Solve all rows of one ALS half-iteration.
 
     explicit: (sum y y^T + lam*nnz*I) x = sum r*y      (ALS-WR, like MLlib)
     implicit: (YtY + sum alpha*r y y^T + lam*I) x = sum (1+alpha*r) y
               (Hu-Koren; YtY required)
+
+    On GPU, rows with nnz <= 32 take the Woodbury path (exact — same
+    linear system solved through the push-through identity); pass a
+    precomputed W = woodbury_w(Y, YtY, lam) to amortize it across calls,
+    else it is computed here for implicit mode.
     """
     if implicit:
         wr_scale = False  # Hu-Koren regularizes with plain lambda*I
@@ -50,10 +67,15 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
         if YtY is not None:
             YtYp = (YtY if pf == f
                     else torch.nn.functional.pad(YtY, (0, pf - f, 0, pf - f)))
+        if implicit and W is None and pf <= 64:
+            W = woodbury_w(Yp, YtYp, lam)
+        elif W is not None and pf != f:
+            W = torch.nn.functional.pad(W, (0, pf - f))
         X = hip_ext().als_solve(
             indptr.contiguous(), indices.contiguous(), values.contiguous(),
             Yp.contiguous(),
             YtYp.contiguous() if YtYp is not None else None,
+            W.contiguous() if W is not None else None,
             float(lam), float(alpha), bool(implicit), bool(wr_scale))
         return X[:, :f].contiguous() if pf != f else X
     return als_solve_ref(indptr, indices, values, Y, YtY, lam, alpha,

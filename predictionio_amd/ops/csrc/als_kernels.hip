@@ -38,7 +38,8 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
     float lambda,
     float alpha,
     int implicit_mode,                      // 1 = Hu-Koren implicit
-    int wr_scale)                           // 1 = scale lambda by nnz (ALS-WR)
+    int wr_scale,                           // 1 = scale lambda by nnz (ALS-WR)
+    int skip_below)                         // skip rows with nnz <= this
 {
   constexpr int TM = 4;                       // thread tile edge
   constexpr int TILES = (F / TM) * (F / TM);  // tiles covering FxF
@@ -60,6 +61,7 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
     const long long start = indptr[row];
     const long long end = indptr[row + 1];
     const int nnz = (int)(end - start);
+    if (nnz <= skip_below) continue;  // Woodbury kernel owns these rows
 
     // register accumulators: TPT tiles of TMxTM
     float acc[TPT][TM][TM];
@@ -198,21 +200,25 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Wave-per-row solver (F <= 64) — the fast path.
+// Wave-per-row solver (F <= 64) — the fast path for mid-size rows.
 //
 // The workgroup kernel above spends most of its time in __syncthreads-heavy
 // in-LDS Cholesky (measured: iteration time nearly flat in nnz — fixed
 // per-row solve cost dominates). This version assigns one 64-lane wave per
-// output row and keeps the row's Gramian slice in REGISTERS:
+// output row (2 waves per workgroup), keeps the Gramian in REGISTERS during
+// accumulation, and factorizes with the computed L columns persisted in LDS:
 //   - lane j owns row j of the FxF Gramian: acc[F] VGPRs, literal-indexed
 //     via fully unrolled loops (runtime-indexed register arrays spill).
-//   - Cholesky: per step k, the pivot column is shared through a 256 B
-//     per-wave LDS buffer (broadcast reads are conflict-free); no
+//   - Cholesky: per step k the scaled pivot column is written to a per-wave
+//     LDS matrix Lc[k][*] (broadcast reads are conflict-free); no
 //     __syncthreads — waves are free-running, intra-wave LDS ordering is
-//     enforced by data dependence + wave_barrier.
-//   - Back-solve trick: lane k captures column k of L into its dead
-//     upper-triangle registers during the trailing update, so L^T solve is
-//     O(F) shuffles with zero extra VGPRs.
+//     enforced by lockstep execution + wave_barrier. Persisting EVERY column
+//     (rather than capturing the transpose into upper-triangle registers)
+//     keeps register live ranges short: a first cut that captured L^T into
+//     dead acc[] registers ballooned to 256 VGPRs + 3 KB scratch spill and
+//     ran 2x slower than the workgroup kernel (rocprof: private_segment
+//     3020 B).
+//   - back solve reads L^T straight from the persisted LDS columns.
 //   - Gramian: per rated item, each lane loads its element of y (coalesced
 //     256 B line), stages it through LDS, and rank-1-updates its register
 //     row from broadcast float4 reads; the next item's global load is
@@ -224,7 +230,7 @@ __device__ __forceinline__ void wave_sync() {
 }
 
 template <int F>
-__global__ __launch_bounds__(256) void als_solve_wave_kernel(
+__global__ __launch_bounds__(128) void als_solve_wave_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
     const float* __restrict__ values,
@@ -235,19 +241,21 @@ __global__ __launch_bounds__(256) void als_solve_wave_kernel(
     float lambda,
     float alpha,
     int implicit_mode,
-    int wr_scale)
+    int wr_scale,
+    int skip_below)
 {
   static_assert(F <= 64, "wave kernel supports rank <= 64");
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  __shared__ float ys[4][2][F];     // double-buffered staged y per wave
-  __shared__ float col_lds[4][64];  // Cholesky pivot-column broadcast
+  __shared__ float ys[2][2][F];        // double-buffered staged y per wave
+  __shared__ float Lc[2][F][F + 1];    // persisted L columns: Lc[w][k][j] = L[j][k]
 
-  for (long long row = (long long)blockIdx.x * 4 + wave; row < n_rows;
-       row += (long long)gridDim.x * 4) {
+  for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
+       row += (long long)gridDim.x * 2) {
     const long long start = indptr[row];
     const int nnz = (int)(indptr[row + 1] - start);
+    if (nnz <= skip_below) continue;  // Woodbury kernel owns these rows
 
     float acc[F];
 #pragma unroll
@@ -313,45 +321,38 @@ __global__ __launch_bounds__(256) void als_solve_wave_kernel(
         if (m == lane) acc[m] += reg;
     }
 
-    // ---- register Cholesky (fully unrolled; lane j = row j) ----
+    // ---- Cholesky (k-loop unrolled so acc[] stays in registers; each
+    //      finished column goes to LDS and acc[k] dies at step k) ----
 #pragma unroll
     for (int k = 0; k < F; ++k) {
       float lkk = __shfl(acc[k], k);
+      // clamp so a singular (all-zero) row divides by a tiny nonzero pivot
+      // instead of 0 (empty rows must yield x = 0, not NaN)
       lkk = lkk > 1e-30f ? lkk : 1e-30f;
       const float dinv = rsqrtf(lkk);
-      // column scale — lanes < k must NOT touch acc[k]: it already holds a
-      // captured transpose value L[k][lane] from an earlier step
-      if (lane >= k) acc[k] *= dinv;
-      col_lds[wave][lane] = acc[k];
+      const float ljk = lane > k ? acc[k] * dinv
+                                 : (lane == k ? lkk * dinv : 0.f);
+      Lc[wave][k][lane] = ljk;  // persist column k: Lc[k][j] = L[j][k]
       wave_sync();
-      const float ljk = acc[k];
 #pragma unroll
       for (int m = k + 1; m < F; ++m) {
-        const float lmk = col_lds[wave][m];  // broadcast read
-        if (lane > k) {
-          acc[m] = fmaf(-ljk, lmk, acc[m]);  // trailing update (rows > k)
-        } else if (lane == k) {
-          acc[m] = lmk;  // capture column k into dead upper registers
-        }
+        const float lmk = Lc[wave][k][m];  // broadcast read
+        if (lane > k) acc[m] = fmaf(-ljk, lmk, acc[m]);
       }
       wave_sync();
     }
 
-    // ---- forward solve L z = b (z in b_reg) ----
-#pragma unroll
+    // ---- forward solve L z = b (z in b_reg); L read from LDS columns ----
     for (int k = 0; k < F; ++k) {
-      const float lkk = __shfl(acc[k], k);
-      const float zk = __shfl(b_reg, k) / lkk;
+      const float zk = __shfl(b_reg, k) / Lc[wave][k][k];
       if (lane == k) b_reg = zk;
-      else if (lane > k) b_reg = fmaf(-acc[k], zk, b_reg);
+      else if (lane > k) b_reg = fmaf(-Lc[wave][k][lane], zk, b_reg);
     }
-    // ---- back solve L^T x = z: acc[k] (k > lane) holds L[k][lane] ----
-#pragma unroll
+    // ---- back solve L^T x = z: lane j reads L[k][j] = Lc[k][j] ----
     for (int k = F - 1; k >= 0; --k) {
-      const float lkk = __shfl(acc[k], k);
-      const float xk = __shfl(b_reg, k) / lkk;
+      const float xk = __shfl(b_reg, k) / Lc[wave][k][k];
       if (lane == k) b_reg = xk;
-      else if (lane < k) b_reg = fmaf(-acc[k], xk, b_reg);
+      else if (lane < k) b_reg = fmaf(-Lc[wave][k][lane], xk, b_reg);
     }
 
     if (lane < F) X[row * (long long)F + lane] = b_reg;
@@ -359,38 +360,210 @@ __global__ __launch_bounds__(256) void als_solve_wave_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Woodbury solver — rows with nnz <= WOODBURY_MAX_NNZ (the common case for
+// recommendation data: the bench config has nnz/row = 20, rank = 64).
+//
+// Instead of building + factorizing the FxF normal matrix per row, exploit
+// that only nnz rank-1 terms differ from a FIXED base B:
+//   implicit (Hu-Koren):  A_u = B + U C U^T,  B = Y^T Y + lambda I,
+//     U = [y_i] (F x n), C = diag(alpha r_i), b_u = U c2, c2_i = 1+alpha r_i.
+//     Push-through identity:  x_u = W_u (I + C G)^-1 c2,
+//     where W = B^-1 Y is precomputed ONCE per half-iteration (host GEMM)
+//     and G = U^T B^-1 U = Y_u . W_u^T (n x n, symmetric).
+//     Symmetrized with D = C^(1/2):  solve (I + D G D) t = c2 / d, s = d t,
+//     x = sum_i s_i w_i.   (I + DGD is SPD.)
+//   explicit (ALS-WR):  A_u = U U^T + reg I, b_u = U r:
+//     x_u = U (G + reg I)^-1 r with G = Y_u . Y_u^T  (W never needed).
+//
+// Cost per row: n^2 F MACs for G + n^3/3 solve — ~5x fewer FLOPs than the
+// FxF Cholesky at n=20, F=64, with NO long serial dependency chains.
+// One wave per row, 2 waves per workgroup; rows with nnz > WOODBURY_MAX_NNZ
+// are skipped here and handled by the workgroup kernel (skip_small=1).
+// LDS rows padded to F+4 floats so G-dot reads from row i (stride 68) land
+// in different bank groups per lane.
+// ---------------------------------------------------------------------------
+
+#define WOODBURY_MAX_NNZ 32
+
+template <int F>
+__global__ __launch_bounds__(128) void als_woodbury_kernel(
+    const long long* __restrict__ indptr,
+    const int* __restrict__ indices,
+    const float* __restrict__ values,
+    const float* __restrict__ Y,
+    const float* __restrict__ W,     // B^-1 Y (implicit mode), else nullptr
+    float* __restrict__ X,
+    int n_rows,
+    float lambda,
+    float alpha,
+    int implicit_mode,
+    int wr_scale)
+{
+  constexpr int NW = WOODBURY_MAX_NNZ;
+  constexpr int FP = F + 4;  // bank-group padding
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+
+  __shared__ float Yl[2][NW][FP];
+  __shared__ float Wl[2][NW][FP];
+  __shared__ float M[2][NW][NW + 1];   // I+DGD (implicit) / G+regI (explicit)
+  __shared__ float tv[2][NW];          // rhs, then solution t
+  __shared__ float dv[2][NW];          // D diagonal (implicit)
+
+  float* yl = &Yl[wave][0][0];
+  float* wl = implicit_mode ? &Wl[wave][0][0] : &Yl[wave][0][0];
+
+  for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
+       row += (long long)gridDim.x * 2) {
+    const long long start = indptr[row];
+    const int n = (int)(indptr[row + 1] - start);
+    if (n > NW) continue;  // workgroup kernel handles these
+    if (n == 0) {
+      if (lane < F) X[row * (long long)F + lane] = 0.f;
+      continue;
+    }
+
+    // ---- stage factor rows + per-item weights ----
+    for (int c = 0; c < n; ++c) {
+      const long long col = indices[start + c];
+      if (lane < F) {
+        yl[c * FP + lane] = Y[col * F + lane];
+        if (implicit_mode) Wl[wave][c][lane] = W[col * F + lane];
+      }
+    }
+    if (lane < n) {
+      const float r = values[start + lane];
+      if (implicit_mode) {
+        // d = sqrt(alpha r) clamped: items with confidence ~0 still carry
+        // their b-contribution (limit s_i -> c2_i as d -> 0)
+        const float ar = alpha * r;
+        const float d = sqrtf(ar > 1e-12f ? ar : 1e-12f);
+        dv[wave][lane] = d;
+        tv[wave][lane] = (1.f + ar) / d;     // rhs = c2 / d
+      } else {
+        tv[wave][lane] = r;                  // rhs = ratings
+      }
+    }
+    wave_sync();
+
+    // ---- M = I + D G D  (implicit)  or  G + reg I  (explicit) ----
+    const float reg = wr_scale ? lambda * (float)n : lambda;
+    const int nn = n * n;
+    for (int p = lane; p < nn; p += 64) {
+      const int i = p / n;
+      const int j = p - i * n;
+      float dot = 0.f;
+      const float* yi = &yl[i * FP];
+      const float* wj = &wl[j * FP];
+#pragma unroll
+      for (int q = 0; q < F / 4; ++q) {
+        const float4 a4 = *reinterpret_cast<const float4*>(yi + 4 * q);
+        const float4 b4 = *reinterpret_cast<const float4*>(wj + 4 * q);
+        dot = fmaf(a4.x, b4.x, dot);
+        dot = fmaf(a4.y, b4.y, dot);
+        dot = fmaf(a4.z, b4.z, dot);
+        dot = fmaf(a4.w, b4.w, dot);
+      }
+      if (implicit_mode) {
+        dot *= dv[wave][i] * dv[wave][j];
+        if (i == j) dot += 1.f;
+      } else if (i == j) {
+        dot += reg;
+      }
+      M[wave][i][j] = dot;
+    }
+    wave_sync();
+
+    // ---- in-LDS Cholesky of M (n x n, lanes 0..n-1 = rows) ----
+    for (int k = 0; k < n; ++k) {
+      float mkk = M[wave][k][k];
+      mkk = mkk > 1e-30f ? mkk : 1e-30f;
+      const float dinv = rsqrtf(mkk);
+      if (lane == k) M[wave][k][k] = mkk * dinv;
+      else if (lane > k && lane < n) M[wave][lane][k] *= dinv;
+      wave_sync();
+      if (lane > k && lane < n) {
+        const float lik = M[wave][lane][k];
+        for (int j = k + 1; j <= lane; ++j)
+          M[wave][lane][j] = fmaf(-lik, M[wave][j][k], M[wave][lane][j]);
+      }
+      wave_sync();
+    }
+    // forward + back substitution on tv (lane-parallel updates)
+    for (int k = 0; k < n; ++k) {
+      if (lane == k) tv[wave][k] /= M[wave][k][k];
+      wave_sync();
+      const float zk = tv[wave][k];
+      if (lane > k && lane < n)
+        tv[wave][lane] = fmaf(-M[wave][lane][k], zk, tv[wave][lane]);
+      wave_sync();
+    }
+    for (int k = n - 1; k >= 0; --k) {
+      if (lane == k) tv[wave][k] /= M[wave][k][k];
+      wave_sync();
+      const float xk = tv[wave][k];
+      if (lane < k)
+        tv[wave][lane] = fmaf(-M[wave][k][lane], xk, tv[wave][lane]);
+      wave_sync();
+    }
+    if (implicit_mode && lane < n) tv[wave][lane] *= dv[wave][lane];  // s = d t
+    wave_sync();
+
+    // ---- x = sum_i s_i w_i  (implicit)  or  sum_i s_i y_i  (explicit) ----
+    if (lane < F) {
+      float x = 0.f;
+      for (int c = 0; c < n; ++c)
+        x = fmaf(tv[wave][c], wl[c * FP + lane], x);
+      X[row * (long long)F + lane] = x;
+    }
+    wave_sync();
+  }
+}
+
+// Variant flag for als_solve_kernel: skip rows the Woodbury kernel owns.
+
+// ---------------------------------------------------------------------------
 // launcher
 // ---------------------------------------------------------------------------
 
 extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
-    const float* Y, const float* YtY, float* X,
+    const float* Y, const float* YtY, const float* W, float* X,
     int n_rows, int f, float lambda, float alpha,
     int implicit_mode, int wr_scale, hipStream_t stream)
 {
-  // >> 256 workgroups to fill 256 CUs across 8 XCDs; one WG per row with
-  // grid-stride for huge row counts.
+  // >> 256 workgroups to fill 256 CUs across 8 XCDs; grid-stride for huge
+  // row counts. W != nullptr (implicit) or explicit mode with f <= 64
+  // enables the Woodbury fast path for rows with nnz <= WOODBURY_MAX_NNZ;
+  // the dense-Gramian kernels then skip those rows.
   if (n_rows <= 0) return;
-  dim3 block(256);
-  // wave kernel: 4 rows per workgroup
-  long long wg = ((long long)n_rows + 3) / 4;
+  // wave/woodbury kernels: 2 rows per 128-thread workgroup
+  long long wg = ((long long)n_rows + 1) / 2;
   int grid_w = (int)(wg < (1 << 20) ? wg : (1 << 20));
   int grid_b = n_rows < (1 << 20) ? n_rows : (1 << 20);
-#define LAUNCH_WAVE(FF)                                                    \
-  hipLaunchKernelGGL((als_solve_wave_kernel<FF>), dim3(grid_w), block, 0,  \
-                     stream, indptr, indices, values, Y, YtY, X, n_rows,   \
-                     lambda, alpha, implicit_mode, wr_scale)
-#define LAUNCH_BLOCK(FF)                                                   \
-  hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), block, 0,       \
-                     stream, indptr, indices, values, Y, YtY, X, n_rows,   \
+  const bool woodbury = f <= 64 && (implicit_mode ? W != nullptr : true);
+  const int skip = woodbury ? WOODBURY_MAX_NNZ : -1;
+#define LAUNCH_WAVE(FF)                                                      \
+  hipLaunchKernelGGL((als_solve_wave_kernel<FF>), dim3(grid_w), dim3(128),   \
+                     0, stream, indptr, indices, values, Y, YtY, X, n_rows,  \
+                     lambda, alpha, implicit_mode, wr_scale, skip)
+#define LAUNCH_BLOCK(FF)                                                     \
+  hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,     \
+                     stream, indptr, indices, values, Y, YtY, X, n_rows,     \
+                     lambda, alpha, implicit_mode, wr_scale, skip)
+#define LAUNCH_WOODBURY(FF)                                                  \
+  if (woodbury)                                                              \
+  hipLaunchKernelGGL((als_woodbury_kernel<FF>), dim3(grid_w), dim3(128),     \
+                     0, stream, indptr, indices, values, Y, W, X, n_rows,    \
                      lambda, alpha, implicit_mode, wr_scale)
   switch (f) {
-    case 16: LAUNCH_WAVE(16); break;
-    case 32: LAUNCH_WAVE(32); break;
-    case 64: LAUNCH_WAVE(64); break;
+    case 16: LAUNCH_WOODBURY(16); LAUNCH_WAVE(16); break;
+    case 32: LAUNCH_WOODBURY(32); LAUNCH_WAVE(32); break;
+    case 64: LAUNCH_WOODBURY(64); LAUNCH_WAVE(64); break;
     case 128: LAUNCH_BLOCK(128); break;
     default: break;  // caller validates
   }
 #undef LAUNCH_WAVE
 #undef LAUNCH_BLOCK
+#undef LAUNCH_WOODBURY
 }

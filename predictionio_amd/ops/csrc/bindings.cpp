@@ -12,7 +12,7 @@
 
 extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
-    const float* Y, const float* YtY, float* X,
+    const float* Y, const float* YtY, const float* W, float* X,
     int n_rows, int f, float lambda, float alpha,
     int implicit_mode, int wr_scale, hipStream_t stream);
 
@@ -39,9 +39,13 @@ bool supported_rank(int64_t f) {
 //   explicit: (sum_i y_i y_i^T + lambda*nnz_r*I) x_r = sum_i r_i y_i
 //   implicit: (YtY + sum_i alpha*r * y_i y_i^T + lambda*I) x_r
 //               = sum_i (1+alpha*r) y_i
+// W = (YtY + lambda I)^-1 Y precomputed on the host side per half-iteration
+// (torch cholesky_solve); enables the per-row Woodbury fast path for small
+// nnz in implicit mode (see als_woodbury_kernel).
 torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
                         torch::Tensor values, torch::Tensor Y,
                         c10::optional<torch::Tensor> YtY,
+                        c10::optional<torch::Tensor> W,
                         double lambda, double alpha,
                         bool implicit_mode, bool wr_scale) {
   TORCH_CHECK(indptr.is_cuda() && indptr.scalar_type() == torch::kInt64 &&
@@ -60,12 +64,18 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
     TORCH_CHECK(YtY->size(0) == f && YtY->size(1) == f, "YtY must be f x f");
     yty_ptr = YtY->data_ptr<float>();
   }
+  const float* w_ptr = nullptr;
+  if (W.has_value()) {
+    check_cuda_f32(*W, "W");
+    TORCH_CHECK(W->sizes() == Y.sizes(), "W must match Y shape");
+    w_ptr = W->data_ptr<float>();
+  }
   auto X = torch::empty({n_rows, f}, Y.options());
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   launch_als_solve(reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()), indices.data_ptr<int>(),
                    values.data_ptr<float>(), Y.data_ptr<float>(), yty_ptr,
-                   X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
+                   w_ptr, X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
                    (float)alpha, implicit_mode ? 1 : 0, wr_scale ? 1 : 0,
                    stream);
   C10_HIP_CHECK(hipGetLastError());
@@ -127,7 +137,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X (gfx950) HIP kernels for predictionio_amd";
   m.def("als_solve", &als_solve, "Fused ALS Gramian+Cholesky half-iteration",
         py::arg("indptr"), py::arg("indices"), py::arg("values"),
-        py::arg("Y"), py::arg("YtY") = py::none(), py::arg("lambda_") = 0.01,
+        py::arg("Y"), py::arg("YtY") = py::none(), py::arg("W") = py::none(),
+        py::arg("lambda_") = 0.01,
         py::arg("alpha") = 1.0, py::arg("implicit_mode") = false,
         py::arg("wr_scale") = true);
   m.def("topk_score", &topk_score, "Fused masked top-K scoring",

@@ -6,8 +6,11 @@ fixed (solve) cost or the streaming (Gramian) cost dominates, and prints
 achieved rows/s + effective HBM GB/s for the Gramian read stream.
 """
 import math
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -57,6 +57,34 @@ class TestALSKernel:
         assert torch.allclose(X, X_ref, atol=2e-3, rtol=2e-3)
         assert X[0].abs().max().item() == 0.0
 
+    @pytest.mark.parametrize("implicit", [False, True])
+    def test_woodbury_seam(self, implicit):
+        """Rows straddling WOODBURY_MAX_NNZ=32: small rows take the Woodbury
+        path, large rows the dense-Gramian path — both must match the
+        reference and the skip logic must leave no row unwritten."""
+        from predictionio_amd.ops import als as als_ops
+        g = torch.Generator().manual_seed(99)
+        f, n_cols = 64, 400
+        sizes = [0, 1, 5, 20, 31, 32, 33, 40, 64, 100, 2, 32, 33, 0, 7]
+        rows, cols, vals = [], [], []
+        for r, n in enumerate(sizes):
+            rows += [r] * n
+            cols.append(torch.randint(0, n_cols, (n,), generator=g))
+            vals.append(torch.rand(n, generator=g) * 3 + 0.5)
+        indptr = torch.tensor([0] + list(torch.tensor(sizes).cumsum(0)),
+                              dtype=torch.int64)
+        cols = torch.cat(cols).to(torch.int32)
+        vals = torch.cat(vals).float()
+        Y = (torch.randn((n_cols, f), generator=g) / math.sqrt(f)).float()
+        YtY = als_ops.gramian(Y) if implicit else None
+        X_ref = als_ops.als_solve_ref(indptr, cols, vals, Y, YtY=YtY,
+                                      lam=0.05, alpha=2.0, implicit=implicit)
+        X = als_ops.als_solve(indptr.cuda(), cols.cuda(), vals.cuda(),
+                              Y.cuda(), lam=0.05, alpha=2.0,
+                              implicit=implicit).cpu()
+        assert torch.allclose(X, X_ref, atol=3e-3, rtol=3e-3), \
+            f"max abs diff {(X - X_ref).abs().max().item()}"
+
     def test_large_row(self):
         """A row with nnz >> chunk size exercises the staging loop."""
         from predictionio_amd.ops import als as als_ops
