@@ -1,0 +1,265 @@
+"""Engine (query) server — serves /queries.json for a deployed engine.
+
+Parity with the reference prediction server (core/.../workflow/
+CreateServer.scala):
+- loads the latest COMPLETED engine instance's models and engine params
+  (createPredictionServerWithEngine :193-250, engineInstanceToEngineParams)
+- POST /queries.json: JSON → Query → serving.supplement → per-algorithm
+  predict (sequential, :506-513) → serving.serve → JSON (:484-634)
+- feedback loop: posts a `predict` event with prId back to the Event
+  Server when feedback is enabled (:527-589)
+- GET /: status page with engine info + requestCount / avgServingSec /
+  lastServingSec counters (:415-417, :597-604)
+- GET /reload: hot-swap to the latest completed instance (:342-371)
+- POST /stop: key-authenticated shutdown (:635-652)
+- GET /plugins.json (:656-678); outputblocker/outputsniffer SPI
+  (EngineServerPlugin.scala:24-41)
+
+The reference binds port 8000 via spark-submitted Akka-HTTP; here FastAPI
+on the same port with the same wire contract.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import HTMLResponse, JSONResponse
+
+logger = logging.getLogger(__name__)
+
+
+class EngineServerPlugin:
+    """Output blocker / sniffer SPI (EngineServerPlugin.scala:24-41)."""
+
+    outputblocker = "outputblocker"
+    outputsniffer = "outputsniffer"
+
+    plugin_name = "plugin"
+    plugin_description = ""
+    plugin_type = outputsniffer
+
+    def process(self, engine_instance, query: Any, prediction: Any) -> Any:
+        """Blockers return a (possibly modified) prediction; sniffers
+        observe. Raise to reject."""
+        return prediction
+
+    def handle_rest(self, arguments: dict) -> Any:
+        return {}
+
+
+@dataclass
+class ServerConfig:
+    engine_factory: str
+    engine_variant: str = "engine.json"
+    ip: str = "0.0.0.0"
+    port: int = 8000
+    feedback: bool = False
+    event_server_uri: str = "http://localhost:7070"
+    access_key: Optional[str] = None
+    app_name: Optional[str] = None
+    engine_instance_id: Optional[str] = None  # None = latest completed
+
+
+@dataclass
+class _ServingState:
+    engine: Any
+    engine_params: Any
+    models: List[Any]
+    instance: Any
+    serving: Any
+    algorithms: List[Any]
+    request_count: int = 0
+    avg_serving_sec: float = 0.0
+    last_serving_sec: float = 0.0
+    start_time: float = field(default_factory=time.time)
+
+
+def _load_state(config: ServerConfig):
+    """Resolve instance → engine params → models (prepareDeploy path)."""
+    from predictionio_amd.controller.engine import get_engine
+    from predictionio_amd.data import storage
+
+    engine = get_engine(config.engine_factory)
+    instances = storage.get_meta_data_engine_instances()
+    if config.engine_instance_id:
+        inst = instances.get(config.engine_instance_id)
+    else:
+        # latest COMPLETED instance of this engine factory (+variant if
+        # one was given) — commands/Engine.deploy resolution (:208-245)
+        cands = [i for i in instances.get_all()
+                 if i.status == "COMPLETED"
+                 and i.engine_factory == config.engine_factory
+                 and (config.engine_variant in ("", "engine.json")
+                      or i.engine_variant == config.engine_variant)]
+        cands.sort(key=lambda i: i.start_time)
+        inst = cands[-1] if cands else None
+    if inst is None:
+        raise RuntimeError(
+            f"No COMPLETED engine instance found for "
+            f"{config.engine_factory} variant {config.engine_variant}. "
+            "Run `pio train` first.")
+    ep = engine.engine_instance_to_engine_params(inst)
+    blob = storage.get_model_data_models().get(inst.id)
+    models = engine.prepare_deploy(ep, inst.id,
+                                   blob.models if blob else None)
+    return _ServingState(
+        engine=engine, engine_params=ep, models=models, instance=inst,
+        serving=engine._serving(ep), algorithms=engine._algorithms(ep))
+
+
+def create_app(config: ServerConfig,
+               plugins: Optional[List[EngineServerPlugin]] = None,
+               state: Optional[_ServingState] = None) -> FastAPI:
+    app = FastAPI(title="PredictionIO-AMD Engine Server")
+    plugins = plugins or []
+    blockers = [p for p in plugins
+                if p.plugin_type == EngineServerPlugin.outputblocker]
+    sniffers = [p for p in plugins
+                if p.plugin_type == EngineServerPlugin.outputsniffer]
+    st = state if state is not None else _load_state(config)
+    holder = {"st": st}
+    lock = threading.Lock()
+
+    def _post_feedback(query_json: dict, prediction_json: dict,
+                       pr_id: str) -> None:
+        """Async predict-event feedback (CreateServer.scala:527-589)."""
+        import urllib.request
+        ev = {
+            "event": "predict",
+            "entityType": "pio_pr",
+            "entityId": pr_id,
+            "properties": {"query": query_json,
+                           "prediction": prediction_json},
+        }
+        url = (f"{config.event_server_uri}/events.json"
+               f"?accessKey={config.access_key}")
+        try:
+            req = urllib.request.Request(
+                url, data=json.dumps(ev).encode(),
+                headers={"Content-Type": "application/json"})
+            urllib.request.urlopen(req, timeout=5)
+        except Exception:
+            logger.exception("feedback loop POST failed")
+
+    @app.get("/", response_class=HTMLResponse)
+    def index():
+        s = holder["st"]
+        return f"""<html><head><title>PredictionIO-AMD Engine Server</title>
+</head><body>
+<h1>Engine Server</h1>
+<p>engineFactory: {config.engine_factory}</p>
+<p>engineInstanceId: {s.instance.id}</p>
+<p>requestCount: {s.request_count}</p>
+<p>avgServingSec: {s.avg_serving_sec:.6f}</p>
+<p>lastServingSec: {s.last_serving_sec:.6f}</p>
+</body></html>"""
+
+    @app.get("/status.json")
+    def status():
+        s = holder["st"]
+        return {
+            "engineFactory": config.engine_factory,
+            "engineInstanceId": s.instance.id,
+            "requestCount": s.request_count,
+            "avgServingSec": s.avg_serving_sec,
+            "lastServingSec": s.last_serving_sec,
+            "startTime": s.start_time,
+        }
+
+    @app.get("/plugins.json")
+    def list_plugins():
+        return {"plugins": {
+            p.plugin_name: {"name": p.plugin_name,
+                            "description": p.plugin_description,
+                            "class": type(p).__name__}
+            for p in plugins
+        }}
+
+    @app.post("/queries.json")
+    async def queries(request: Request):
+        s = holder["st"]
+        t0 = time.time()
+        try:
+            query_json = await request.json()
+        except Exception:
+            return JSONResponse({"message": "invalid JSON"},
+                                status_code=400)
+        try:
+            query = s.algorithms[0].query_from_json(query_json) \
+                if hasattr(s.algorithms[0], "query_from_json") \
+                else query_json
+            supplemented = s.serving.supplement(query)
+            predictions = [a.predict(m, supplemented)
+                           for a, m in zip(s.algorithms, s.models)]
+            prediction = s.serving.serve(query, predictions)
+        except Exception as e:
+            logger.exception("query failed")
+            return JSONResponse({"message": str(e)}, status_code=500)
+        for b in blockers:
+            try:
+                prediction = b.process(s.instance, query, prediction)
+            except Exception as e:
+                return JSONResponse({"message": str(e)}, status_code=403)
+        pred_json = (prediction.to_json()
+                     if hasattr(prediction, "to_json") else prediction)
+        if config.feedback and config.access_key:
+            pr_id = pred_json.get("prId") if isinstance(pred_json, dict) \
+                else None
+            pr_id = pr_id or f"pr-{int(t0 * 1000)}"
+            if isinstance(pred_json, dict):
+                pred_json = {**pred_json, "prId": pr_id}
+            threading.Thread(
+                target=_post_feedback,
+                args=(query_json, pred_json, pr_id), daemon=True).start()
+        dt = time.time() - t0
+        with lock:
+            s.request_count += 1
+            s.last_serving_sec = dt
+            s.avg_serving_sec += (dt - s.avg_serving_sec) / s.request_count
+        for sn in sniffers:
+            try:
+                sn.process(s.instance, query, prediction)
+            except Exception:
+                logger.exception("sniffer plugin failed")
+        return pred_json
+
+    @app.get("/reload")
+    def reload():
+        """Hot-swap to the latest completed instance
+        (CreateServer.scala:342-371)."""
+        try:
+            holder["st"] = _load_state(config)
+        except Exception as e:
+            return JSONResponse({"message": str(e)}, status_code=500)
+        return {"message": "Reloaded",
+                "engineInstanceId": holder["st"].instance.id}
+
+    @app.post("/stop")
+    def stop(request: Request):
+        """Key-authenticated shutdown (CreateServer.scala:635-652)."""
+        key = request.query_params.get("accessKey")
+        if config.access_key and key != config.access_key:
+            return JSONResponse({"message": "Invalid accessKey."},
+                                status_code=401)
+        # uvicorn exits when the process receives SIGTERM
+        import signal
+        threading.Timer(
+            0.2, lambda: os.kill(os.getpid(), signal.SIGTERM)).start()
+        return {"message": "Shutting down."}
+
+    return app
+
+
+def run(config: ServerConfig,
+        plugins: Optional[List[EngineServerPlugin]] = None) -> None:
+    """`pio deploy` entry point (reference default port 8000)."""
+    import uvicorn
+    uvicorn.run(create_app(config, plugins), host=config.ip,
+                port=config.port, log_level="info")

@@ -136,6 +136,28 @@ class FakeEngineFactory(EngineFactory):
         )
 
 
+class JsonAlgo(Algorithm):
+    """JSON-facing algorithm: dict queries in, dict predictions out
+    (the wire contract of /queries.json)."""
+
+    def train(self, pd):
+        return sum(pd.xs) * pd.scale
+
+    def predict(self, model, q):
+        return {"result": model + q["x"]}
+
+
+class JsonServing(Serving):
+    def serve(self, q, preds):
+        return {"result": sum(p["result"] for p in preds)}
+
+
+class JsonEngineFactory(EngineFactory):
+    @classmethod
+    def apply(cls):
+        return Engine(DS0, Prep0, JsonAlgo, JsonServing)
+
+
 def make_engine() -> Engine:
     return FakeEngineFactory.apply()
 

@@ -1,0 +1,128 @@
+"""Engine (query) server tests — mirrors the role of the reference's
+CreateServer route behavior (:484-705): query flow, counters, reload,
+plugins, missing-instance error."""
+
+import pytest
+from fastapi.testclient import TestClient
+
+from predictionio_amd.workflow import train as train_wf
+
+FACTORY = "tests.fake_engine.JsonEngineFactory"
+
+
+def _train(variant_id="v1", n=4, scale=1):
+    variant = {
+        "id": variant_id,
+        "engineFactory": FACTORY,
+        "datasource": {"params": {"n": n}},
+        "preparator": {"params": {"scale": scale}},
+        "algorithms": [{"name": "", "params": {}}],
+    }
+    return train_wf.run_train_from_variant(variant)
+
+
+@pytest.fixture()
+def server(mem_storage):
+    _train()
+    from predictionio_amd.server.queryserver import ServerConfig, create_app
+    cfg = ServerConfig(engine_factory=FACTORY, engine_variant="v1")
+    return TestClient(create_app(cfg))
+
+
+class TestQueries:
+    def test_query(self, server):
+        r = server.post("/queries.json", json={"x": 5})
+        assert r.status_code == 200
+        assert r.json() == {"result": 6 + 5}  # sum(0..3)=6
+
+    def test_counters(self, server):
+        for x in range(3):
+            server.post("/queries.json", json={"x": x})
+        s = server.get("/status.json").json()
+        assert s["requestCount"] == 3
+        assert s["avgServingSec"] > 0
+
+    def test_index_page(self, server):
+        r = server.get("/")
+        assert "requestCount" in r.text
+
+    def test_invalid_json(self, server):
+        r = server.post("/queries.json", content=b"not json",
+                        headers={"Content-Type": "application/json"})
+        assert r.status_code == 400
+
+    def test_reload_picks_latest(self, mem_storage):
+        _train(n=4)
+        from predictionio_amd.server.queryserver import (
+            ServerConfig, create_app,
+        )
+        cfg = ServerConfig(engine_factory=FACTORY)
+        c = TestClient(create_app(cfg))
+        assert c.post("/queries.json", json={"x": 0}).json() == {"result": 6}
+        _train(n=5)  # sum(0..4)=10
+        r = c.get("/reload")
+        assert r.status_code == 200
+        assert c.post("/queries.json", json={"x": 0}).json() == {"result": 10}
+
+    def test_no_completed_instance(self, mem_storage):
+        from predictionio_amd.server.queryserver import (
+            ServerConfig, _load_state,
+        )
+        with pytest.raises(RuntimeError, match="No COMPLETED"):
+            _load_state(ServerConfig(engine_factory=FACTORY))
+
+
+class TestPlugins:
+    def test_output_blocker(self, mem_storage):
+        _train()
+        from predictionio_amd.server.queryserver import (
+            EngineServerPlugin, ServerConfig, create_app,
+        )
+
+        class Cap(EngineServerPlugin):
+            plugin_name = "cap"
+            plugin_type = EngineServerPlugin.outputblocker
+
+            def process(self, inst, q, p):
+                return {"result": min(p["result"], 7)}
+
+        c = TestClient(create_app(ServerConfig(engine_factory=FACTORY),
+                                  plugins=[Cap()]))
+        assert c.post("/queries.json", json={"x": 100}).json() == \
+            {"result": 7}
+
+    def test_plugins_json(self, server):
+        assert server.get("/plugins.json").json() == {"plugins": {}}
+
+
+class TestFeedback:
+    def test_feedback_posts_event(self, mem_storage, monkeypatch):
+        _train()
+        from predictionio_amd.server import queryserver as qs
+        posted = []
+        monkeypatch.setattr(
+            qs, "_load_state", qs._load_state)  # keep real loader
+        cfg = qs.ServerConfig(engine_factory=FACTORY, feedback=True,
+                              access_key="K")
+        app = qs.create_app(cfg)
+        c = TestClient(app)
+
+        import urllib.request
+
+        def fake_urlopen(req, timeout=None):
+            posted.append(req)
+            class R:
+                def read(self):
+                    return b"{}"
+            return R()
+
+        monkeypatch.setattr(urllib.request, "urlopen", fake_urlopen)
+        r = c.post("/queries.json", json={"x": 1})
+        assert r.status_code == 200
+        assert "prId" in r.json()
+        import time
+        for _ in range(50):
+            if posted:
+                break
+            time.sleep(0.05)
+        assert posted and "/events.json" in posted[0].full_url
