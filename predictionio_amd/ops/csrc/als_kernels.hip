@@ -348,11 +348,13 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
       if (lane == k) b_reg = zk;
       else if (lane > k) b_reg = fmaf(-Lc[wave][k][lane], zk, b_reg);
     }
-    // ---- back solve L^T x = z: lane j reads L[k][j] = Lc[k][j] ----
+    // ---- back solve L^T x = z: lane j reads L[k][j] = Lc[j][k]
+    //      (Lc[c][r] stores L[r][c]; stride F+1 across lanes → no bank
+    //      conflicts) ----
     for (int k = F - 1; k >= 0; --k) {
       const float xk = __shfl(b_reg, k) / Lc[wave][k][k];
       if (lane == k) b_reg = xk;
-      else if (lane < k) b_reg = fmaf(-Lc[wave][k][lane], xk, b_reg);
+      else if (lane < k) b_reg = fmaf(-Lc[wave][lane][k], xk, b_reg);
     }
 
     if (lane < F) X[row * (long long)F + lane] = b_reg;
@@ -447,30 +449,41 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     wave_sync();
 
     // ---- M = I + D G D  (implicit)  or  G + reg I  (explicit) ----
+    // G is symmetric (G_ij = y_i^T B^-1 y_j): compute the upper triangle
+    // only, mirror on write. Pair index advances incrementally — no
+    // per-iteration integer division.
     const float reg = wr_scale ? lambda * (float)n : lambda;
-    const int nn = n * n;
-    for (int p = lane; p < nn; p += 64) {
-      const int i = p / n;
-      const int j = p - i * n;
-      float dot = 0.f;
-      const float* yi = &yl[i * FP];
-      const float* wj = &wl[j * FP];
+    const int npairs = n * (n + 1) / 2;
+    {
+      // map lane → first (i, j) with j >= i in the flattened triangle
+      int p = lane, i = 0;
+      while (p >= n - i && i < n) { p -= n - i; ++i; }
+      int j = i + p;
+      for (int pp = lane; pp < npairs; pp += 64) {
+        float dot = 0.f;
+        const float* yi = &yl[i * FP];
+        const float* wj = &wl[j * FP];
 #pragma unroll
-      for (int q = 0; q < F / 4; ++q) {
-        const float4 a4 = *reinterpret_cast<const float4*>(yi + 4 * q);
-        const float4 b4 = *reinterpret_cast<const float4*>(wj + 4 * q);
-        dot = fmaf(a4.x, b4.x, dot);
-        dot = fmaf(a4.y, b4.y, dot);
-        dot = fmaf(a4.z, b4.z, dot);
-        dot = fmaf(a4.w, b4.w, dot);
+        for (int q = 0; q < F / 4; ++q) {
+          const float4 a4 = *reinterpret_cast<const float4*>(yi + 4 * q);
+          const float4 b4 = *reinterpret_cast<const float4*>(wj + 4 * q);
+          dot = fmaf(a4.x, b4.x, dot);
+          dot = fmaf(a4.y, b4.y, dot);
+          dot = fmaf(a4.z, b4.z, dot);
+          dot = fmaf(a4.w, b4.w, dot);
+        }
+        if (implicit_mode) {
+          dot *= dv[wave][i] * dv[wave][j];
+          if (i == j) dot += 1.f;
+        } else if (i == j) {
+          dot += reg;
+        }
+        M[wave][i][j] = dot;
+        M[wave][j][i] = dot;
+        // advance 64 triangle slots
+        j += 64;
+        while (j >= n && i < n) { ++i; j -= n - i; }
       }
-      if (implicit_mode) {
-        dot *= dv[wave][i] * dv[wave][j];
-        if (i == j) dot += 1.f;
-      } else if (i == j) {
-        dot += reg;
-      }
-      M[wave][i][j] = dot;
     }
     wave_sync();
 

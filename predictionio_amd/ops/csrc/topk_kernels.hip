@@ -11,11 +11,22 @@
 // Design (MI355X-first): scores = Xq . Y^T fused with masking and an
 // on-device per-slice top-K reduction, never materializing the B x N score
 // matrix (N up to 10^7). Grid = (item_slices, user_blocks); each 256-thread
-// workgroup stages a 64-item chunk of Y in LDS (coalesced, +1-padded
-// against bank conflicts), computes dots for UPB users per wave, and
-// maintains a per-user running top-K list in LDS with a wave-ballot
-// insertion filter. Phase 2 (merging the per-slice candidates) is a small
-// torch.topk on [B, slices*K].
+// workgroup owns a 64-user x item-slice tile:
+//  - 64-item chunks of Y staged through LDS (coalesced), then TRANSPOSED
+//    into per-lane registers (lane l holds item l's full factor row) —
+//    a v1 of this kernel kept 8 users per workgroup and re-read Y B/8
+//    times from HBM (1.3 TB per 4096-query batch, measured 656 ms); with
+//    64 users per block Y traffic drops 8x and the dot loop is
+//    1 fma + 1 LDS-broadcast per score.
+//  - each wave scores 16 users against the 64 staged items: the k-loop
+//    reads xs[u][k] (same address across lanes → LDS broadcast) against
+//    yreg[k] (literal register index).
+//  - per-user running top-K in LDS with a wave-ballot insertion filter:
+//    lanes beating the user's current K-th best serialize through lane 0,
+//    which also applies the per-user banned-list (binary search, short
+//    sorted lists) before insertion.
+// Phase 2 (merging the per-slice candidates) is a small torch.topk on
+// [B, slices*K] host-side tensors.
 //
 // Masks:
 //  - item_mask: optional uint8[N], 1 = globally banned (e.g. unavailable
@@ -40,7 +51,7 @@ __device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
-template <int F, int UPW>  // UPW = users per wave
+template <int F>
 __global__ __launch_bounds__(256) void topk_score_kernel(
     const float* __restrict__ Xq,        // B x F query vectors
     const float* __restrict__ Y,         // N x F item factors
@@ -51,13 +62,15 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     int* __restrict__ out_idx,           // B x n_slices x K
     int B, long long N, int K, int n_slices, int item_base)
 {
-  constexpr int WAVES = 4;
-  constexpr int UPB = WAVES * UPW;  // users per block
-
-  __shared__ float ys[TK_CHUNK][F + 1];
-  __shared__ float xs[UPB][F + 1];
-  __shared__ float topv[UPB][TK_MAXK];
-  __shared__ int topi[UPB][TK_MAXK];
+  constexpr int UPW = 16;           // users per wave
+  constexpr int UPB = 4 * UPW;      // users per block = 64
+  // dynamic LDS layout: ys[64][F+1] | xs[UPB][F+1] | topv[UPB][K] |
+  // topi[UPB][K]
+  extern __shared__ float lds[];
+  float* ys = lds;                              // 64 x (F+1)
+  float* xs = ys + TK_CHUNK * (F + 1);          // UPB x (F+1)
+  float* topv = xs + UPB * (F + 1);             // UPB x K
+  int* topi = reinterpret_cast<int*>(topv + UPB * K);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -76,11 +89,11 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
   for (int e = tid; e < UPB * F; e += 256) {
     const int u = e / F;
     const int k = e % F;
-    xs[u][k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
+    xs[u * (F + 1) + k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
   }
   for (int e = tid; e < UPB * K; e += 256) {
-    topv[e / K][e % K] = -FLT_MAX;
-    topi[e / K][e % K] = -1;
+    topv[e] = -FLT_MAX;
+    topi[e] = -1;
   }
   __syncthreads();
 
@@ -98,26 +111,35 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     for (int e = tid; e < cn * F; e += 256) {
       const int c = e / F;
       const int k = e % F;
-      ys[c][k] = Y[(base + c) * F + k];
+      ys[c * (F + 1) + k] = Y[(base + c) * F + k];
     }
     __syncthreads();
 
-    // each wave handles its UPW users; lane l scores item l of the chunk
+    // transpose the chunk into registers: lane l = item l of the chunk
+    // (stride F+1 across lanes → conflict-free column reads)
+    float yreg[F];
+    const bool live = lane < cn;
+    {
+      const float* yrow = ys + lane * (F + 1);
+#pragma unroll
+      for (int k = 0; k < F; ++k) yreg[k] = live ? yrow[k] : 0.f;
+    }
+    const long long item = base + lane;
+    const bool open = live &&
+        !(item_mask != nullptr && item_mask[item]);
+
+    // each wave scores its UPW users against the 64 staged items
 #pragma unroll
     for (int uu = 0; uu < UPW; ++uu) {
       const int u = wave * UPW + uu;
       const long long guser = u0 + u;
       float s = -FLT_MAX;
-      const long long item = base + lane;
-      if (lane < cn && guser < B) {
-        bool banned = false;
-        if (item_mask != nullptr && item_mask[item]) banned = true;
-        if (!banned) {
-          float acc = 0.f;
-#pragma unroll 8
-          for (int k = 0; k < F; ++k) acc = fmaf(xs[u][k], ys[lane][k], acc);
-          s = acc;
-        }
+      if (open && guser < B) {
+        const float* xrow = xs + u * (F + 1);
+        float acc = 0.f;
+#pragma unroll
+        for (int k = 0; k < F; ++k) acc = fmaf(xrow[k], yreg[k], acc);
+        s = acc;
       }
       // wave-ballot insertion: only lanes beating the running threshold
       unsigned long long mask = __ballot(s > uth[uu]);
@@ -139,15 +161,17 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
           }
           if (!banned && v > nth) {
             // replace current min of the K-list
+            float* tvu = topv + u * K;
+            int* tiu = topi + u * K;
             int mi = 0;
-            float mv = topv[u][0];
+            float mv = tvu[0];
             for (int q = 1; q < K; ++q)
-              if (topv[u][q] < mv) { mv = topv[u][q]; mi = q; }
-            topv[u][mi] = v;
-            topi[u][mi] = (int)(cand + item_base);
+              if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
+            tvu[mi] = v;
+            tiu[mi] = (int)(cand + item_base);
             // new threshold = K-th best = new min of the list
-            float nm = topv[u][0];
-            for (int q = 1; q < K; ++q) nm = fminf(nm, topv[u][q]);
+            float nm = tvu[0];
+            for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
             nth = nm;
           }
         }
@@ -165,8 +189,8 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     const long long guser = u0 + u;
     if (guser < B) {
       const long long o = (guser * n_slices + slice) * K + q;
-      out_val[o] = topv[u][q];
-      out_idx[o] = topi[u][q];
+      out_val[o] = topv[u * K + q];
+      out_idx[o] = topi[u * K + q];
     }
   }
 }
@@ -178,14 +202,25 @@ extern "C" void launch_topk_score(
     int B, long long N, int f, int K, int n_slices, int item_base,
     hipStream_t stream)
 {
-  constexpr int UPW = 2;
-  const int UPB = 4 * UPW;
+  const int UPB = 64;
   dim3 grid(n_slices, (B + UPB - 1) / UPB);
   dim3 block(256);
-#define LAUNCH(FF)                                                        \
-  hipLaunchKernelGGL((topk_score_kernel<FF, UPW>), grid, block, 0, stream,\
-                     Xq, Y, item_mask, ban_indptr, ban_indices, out_val,  \
-                     out_idx, B, N, K, n_slices, item_base)
+#define LAUNCH(FF)                                                          \
+  do {                                                                      \
+    const size_t lds_bytes =                                                \
+        sizeof(float) * ((TK_CHUNK + UPB) * (FF + 1) + UPB * K) +           \
+        sizeof(int) * UPB * K;                                              \
+    static bool attr_set_##FF = false;                                      \
+    if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
+      hipFuncSetAttribute(                                                  \
+          reinterpret_cast<const void*>(&topk_score_kernel<FF>),            \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);          \
+      attr_set_##FF = true;                                                 \
+    }                                                                       \
+    hipLaunchKernelGGL((topk_score_kernel<FF>), grid, block, lds_bytes,     \
+                       stream, Xq, Y, item_mask, ban_indptr, ban_indices,   \
+                       out_val, out_idx, B, N, K, n_slices, item_base);     \
+  } while (0)
   switch (f) {
     case 16: LAUNCH(16); break;
     case 32: LAUNCH(32); break;
