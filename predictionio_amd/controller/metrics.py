@@ -107,6 +107,30 @@ class MetricEvaluatorResult:
     other_metric_headers: List[str]
     engine_params_scores: List[Tuple[Any, float, List[float]]]
 
+    def summary(self) -> str:
+        """One-line-per-candidate report (MetricEvaluator.scala logs +
+        MetricEvaluatorResult.toOneLiner)."""
+        lines = [f"Metric: {self.metric_header}"]
+        for i, (ep, s, others) in enumerate(self.engine_params_scores):
+            mark = " (best)" if i == self.best_idx else ""
+            extra = "".join(f" {h}={v:.6f}" for h, v in
+                            zip(self.other_metric_headers, others))
+            lines.append(f"  [{i}] score={s:.6f}{extra}{mark}")
+        lines.append(f"Best score: {self.best_score:.6f} "
+                     f"(candidate {self.best_idx})")
+        return "\n".join(lines)
+
+    def to_json(self) -> dict:
+        from dataclasses import asdict, is_dataclass
+        ep = self.best_engine_params
+        return {
+            "bestScore": self.best_score,
+            "bestIdx": self.best_idx,
+            "metric": self.metric_header,
+            "bestEngineParams": asdict(ep) if is_dataclass(ep) else str(ep),
+            "scores": [s for _, s, _ in self.engine_params_scores],
+        }
+
 
 class MetricEvaluator:
     """Ranks candidate EngineParams by the primary metric and optionally

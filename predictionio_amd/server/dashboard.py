@@ -1,0 +1,59 @@
+"""Dashboard — lists completed evaluations (port 9000).
+
+Parity with tools/.../dashboard/Dashboard.scala:44-60, 85-155: an HTML
+index of completed EvaluationInstances, most recent first, with per-
+instance HTML and JSON detail pages.
+"""
+
+from __future__ import annotations
+
+from html import escape
+
+from fastapi import FastAPI
+from fastapi.responses import HTMLResponse, JSONResponse
+
+from predictionio_amd.data import storage
+
+
+def create_app() -> FastAPI:
+    app = FastAPI(title="PredictionIO-AMD Dashboard")
+
+    @app.get("/", response_class=HTMLResponse)
+    def index():
+        insts = storage.get_meta_data_evaluation_instances().get_completed()
+        rows = "".join(
+            f"<tr><td><a href='/engine_instances/{i.id}'>{i.id}</a></td>"
+            f"<td>{escape(i.evaluation_class)}</td>"
+            f"<td>{escape(i.engine_params_generator_class)}</td>"
+            f"<td>{i.start_time}</td><td>{i.end_time}</td>"
+            f"<td><a href='/engine_instances/{i.id}/evaluator_results.json'>"
+            "JSON</a></td></tr>"
+            for i in insts)
+        return f"""<html><head><title>PredictionIO-AMD Dashboard</title>
+</head><body><h1>Completed Evaluations</h1>
+<table border=1 cellpadding=4>
+<tr><th>ID</th><th>Evaluation</th><th>Generator</th><th>Start</th>
+<th>End</th><th>Results</th></tr>{rows}</table></body></html>"""
+
+    @app.get("/engine_instances/{iid}", response_class=HTMLResponse)
+    def detail(iid: str):
+        i = storage.get_meta_data_evaluation_instances().get(iid)
+        if i is None:
+            return HTMLResponse("<h1>Not Found</h1>", status_code=404)
+        return (f"<html><body><h1>Evaluation {i.id}</h1>"
+                f"{i.evaluator_results_html or ''}</body></html>")
+
+    @app.get("/engine_instances/{iid}/evaluator_results.json")
+    def detail_json(iid: str):
+        i = storage.get_meta_data_evaluation_instances().get(iid)
+        if i is None:
+            return JSONResponse({"message": "Not Found"}, status_code=404)
+        import json as _json
+        return _json.loads(i.evaluator_results_json or "{}")
+
+    return app
+
+
+def run(host: str = "127.0.0.1", port: int = 9000) -> None:
+    import uvicorn
+    uvicorn.run(create_app(), host=host, port=port, log_level="info")

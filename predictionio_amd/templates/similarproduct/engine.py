@@ -1,0 +1,309 @@
+"""Similar-Product engine template (MI355X-native).
+
+Parity with examples/scala-parallel-similarproduct/recommended-user →
+actually the base template (examples/scala-parallel-similarproduct/*):
+- Query {items, num, [categories], [whiteList], [blackList]}
+  (src/main/scala/Engine.scala)
+- DataSource aggregates $set user/item properties + view events
+  (DataSource.scala:133 total)
+- ALSAlgorithm (P2L): implicit-style ALS on view events with latest-wins
+  (user, item) dedup (ALSAlgorithm.scala:88-120), productFeatures collected
+  local (:130-143); predict = sum of cosine similarities of the query
+  items' factors vs ALL item factors, filtered by white/black/category,
+  top-num (:168-271 — the reference's .par loop + hand-rolled cosine +
+  bounded PriorityQueue)
+- CooccurrenceAlgorithm: user-distinct item pairs → pair counts → top-N
+  co-occurring items (CooccurrenceAlgorithm.scala:49-108)
+
+MI355X design: cosine scoring = L2-normalize factors once at train time,
+then the summed-query-vector dot against all items in one fused masked
+top-K kernel launch.
+"""
+
+from __future__ import annotations
+
+from collections import defaultdict
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional, Set, Tuple
+
+import torch
+
+from predictionio_amd.controller import (
+    Algorithm, DataSource as BaseDataSource, Engine, EngineFactory,
+    Preparator as BasePreparator, SanityCheck, Serving as BaseServing,
+)
+from predictionio_amd.data import event_store
+from predictionio_amd.data.bimap import BiMap
+from predictionio_amd.models.als import ALSParams, train_als
+from predictionio_amd.ops import topk as topk_ops
+
+
+@dataclass
+class Item:
+    categories: Optional[List[str]]
+
+
+@dataclass
+class ViewEvent:
+    user: str
+    item: str
+    t: float
+
+
+@dataclass
+class TrainingData(SanityCheck):
+    users: Dict[str, dict]
+    items: Dict[str, Item]
+    view_events: List[ViewEvent]
+
+    def sanity_check(self):
+        if not self.view_events:
+            raise ValueError("view events are empty")
+
+
+@dataclass
+class PreparedData:
+    users: Dict[str, dict]
+    items: Dict[str, Item]
+    view_events: List[ViewEvent]
+
+
+@dataclass
+class Query:
+    items: List[str]
+    num: int
+    categories: Optional[List[str]] = None
+    white_list: Optional[List[str]] = None
+    black_list: Optional[List[str]] = None
+
+    @staticmethod
+    def from_json(d: Dict[str, Any]) -> "Query":
+        return Query(items=list(d["items"]), num=int(d.get("num", 10)),
+                     categories=d.get("categories"),
+                     white_list=d.get("whiteList"),
+                     black_list=d.get("blackList"))
+
+
+@dataclass
+class ItemScore:
+    item: str
+    score: float
+
+
+@dataclass
+class PredictedResult:
+    item_scores: List[ItemScore]
+
+    def to_json(self):
+        return {"itemScores": [{"item": s.item, "score": s.score}
+                               for s in self.item_scores]}
+
+
+class DataSource(BaseDataSource):
+    """Params: appName."""
+
+    def read_training(self) -> TrainingData:
+        app = self.params["appName"]
+        users = {eid: pm.to_dict() for eid, pm in
+                 event_store.aggregate_properties(app, "user").items()}
+        items = {eid: Item(categories=pm.get_opt("categories"))
+                 for eid, pm in
+                 event_store.aggregate_properties(app, "item").items()}
+        views = [
+            ViewEvent(e.entity_id, e.target_entity_id,
+                      e.event_time.timestamp())
+            for e in event_store.find(app, entity_type="user",
+                                      event_names=["view"],
+                                      target_entity_type="item")
+        ]
+        return TrainingData(users, items, views)
+
+
+class Preparator(BasePreparator):
+    def prepare(self, td: TrainingData) -> PreparedData:
+        return PreparedData(td.users, td.items, td.view_events)
+
+
+class SimilarModel:
+    """Normalized item factors + maps + categories (P2L local model)."""
+
+    def __init__(self, item_factors_norm: torch.Tensor, item_map: BiMap,
+                 items: Dict[str, Item]):
+        self.item_factors_norm = item_factors_norm  # L2-normalized rows
+        self.item_map = item_map
+        self.items = items
+        self.item_inv = item_map.inverse_array()
+
+
+class ALSAlgorithm(Algorithm):
+    """Params: rank, numIterations, lambda, alpha, [seed]."""
+
+    def train(self, pd: PreparedData) -> SimilarModel:
+        if not pd.view_events:
+            raise ValueError("no view events")
+        user_map = BiMap.string_int(v.user for v in pd.view_events)
+        # include property-only items so queries on them resolve
+        item_keys = [v.item for v in pd.view_events] + list(pd.items)
+        item_map = BiMap.string_int(item_keys)
+        # latest-wins dedup on (user, item) (ALSAlgorithm.scala:88-120):
+        # sort by time then aggregate 'latest'; implicit weight 1 per view
+        evs = sorted(pd.view_events, key=lambda v: v.t)
+        users = torch.tensor([user_map[v.user] for v in evs],
+                             dtype=torch.int32)
+        items = torch.tensor([item_map[v.item] for v in evs],
+                             dtype=torch.int32)
+        vals = torch.ones(len(evs))
+        from predictionio_amd.ops import als as als_ops
+        users, items, vals = als_ops.aggregate_ratings(
+            users, items, vals, len(item_map), "sum")
+        p = ALSParams(
+            rank=int(self.params.get("rank", 10)),
+            iterations=int(self.params.get("numIterations", 20)),
+            lambda_=float(self.params.get("lambda", 0.01)),
+            alpha=float(self.params.get("alpha", 1.0)),
+            implicit=True, seed=self.params.get("seed"))
+        device = torch.device("cuda") if torch.cuda.is_available() \
+            else torch.device("cpu")
+        _, Y = train_als(users, items, vals, len(user_map), len(item_map),
+                         p, device=device)
+        Yn = torch.nn.functional.normalize(Y, dim=1, eps=1e-9)
+        return SimilarModel(Yn, item_map, pd.items)
+
+    def _masks(self, model: SimilarModel, q: Query, dev) -> Optional[torch.Tensor]:
+        """uint8 banned mask from white/black/category filters
+        (ALSAlgorithm.scala:245-271 isCandidateItem)."""
+        n = len(model.item_map)
+        mask = None
+        if q.white_list is not None:
+            mask = torch.ones(n, dtype=torch.uint8)
+            for it in q.white_list:
+                i = model.item_map.get(it)
+                if i is not None:
+                    mask[i] = 0
+        if q.categories is not None:
+            cm = torch.ones(n, dtype=torch.uint8)
+            cats = set(q.categories)
+            for it, meta in model.items.items():
+                if meta.categories and cats & set(meta.categories):
+                    i = model.item_map.get(it)
+                    if i is not None:
+                        cm[i] = 0
+            mask = cm if mask is None else (mask | cm)
+        if q.black_list:
+            if mask is None:
+                mask = torch.zeros(n, dtype=torch.uint8)
+            for it in q.black_list:
+                i = model.item_map.get(it)
+                if i is not None:
+                    mask[i] = 1
+        # query items themselves are never returned (reference excludes them)
+        if mask is None:
+            mask = torch.zeros(n, dtype=torch.uint8)
+        for it in q.items:
+            i = model.item_map.get(it)
+            if i is not None:
+                mask[i] = 1
+        return mask.to(dev)
+
+    def predict(self, model: SimilarModel, query) -> PredictedResult:
+        q = query if isinstance(query, Query) else Query.from_json(query)
+        rows = [model.item_map[i] for i in q.items
+                if i in model.item_map]
+        if not rows:
+            return PredictedResult([])
+        dev = model.item_factors_norm.device
+        # sum of cosines = (sum of normalized query vectors) . Yn
+        Xq = model.item_factors_norm[torch.tensor(rows, device=dev)] \
+            .sum(dim=0, keepdim=True)
+        mask = self._masks(model, q, dev)
+        v, idx = topk_ops.topk_score(Xq, model.item_factors_norm, q.num,
+                                     item_mask=mask)
+        v, idx = v[0].cpu(), idx[0].cpu()
+        return PredictedResult([
+            ItemScore(model.item_inv[int(i)], float(s))
+            for s, i in zip(v, idx) if i >= 0])
+
+
+class CooccurrenceModel:
+    def __init__(self, top_n: Dict[int, List[Tuple[int, int]]],
+                 item_map: BiMap, items: Dict[str, Item]):
+        self.top_n = top_n          # item → [(other_item, count)]
+        self.item_map = item_map
+        self.items = items
+        self.item_inv = item_map.inverse_array()
+
+
+class CooccurrenceAlgorithm(Algorithm):
+    """Params: n (top co-occurrences kept per item)
+    (CooccurrenceAlgorithm.scala:49-108)."""
+
+    def train(self, pd: PreparedData) -> CooccurrenceModel:
+        item_map = BiMap.string_int(
+            [v.item for v in pd.view_events] + list(pd.items))
+        n_keep = int(self.params.get("n", 10))
+        # user-distinct item sets → pair counts
+        by_user: Dict[str, Set[int]] = defaultdict(set)
+        for v in pd.view_events:
+            by_user[v.user].add(item_map[v.item])
+        pair_counts: Dict[Tuple[int, int], int] = defaultdict(int)
+        for items in by_user.values():
+            s = sorted(items)
+            for a_i, a in enumerate(s):
+                for b in s[a_i + 1:]:
+                    pair_counts[(a, b)] += 1
+        per_item: Dict[int, List[Tuple[int, int]]] = defaultdict(list)
+        for (a, b), c in pair_counts.items():
+            per_item[a].append((b, c))
+            per_item[b].append((a, c))
+        top_n = {i: sorted(l, key=lambda t: -t[1])[:n_keep]
+                 for i, l in per_item.items()}
+        return CooccurrenceModel(top_n, item_map, pd.items)
+
+    def predict(self, model: CooccurrenceModel, query) -> PredictedResult:
+        q = query if isinstance(query, Query) else Query.from_json(query)
+        counts: Dict[int, int] = defaultdict(int)
+        qidx = {model.item_map.get(i) for i in q.items}
+        for i in q.items:
+            ii = model.item_map.get(i)
+            if ii is None:
+                continue
+            for other, c in model.top_n.get(ii, []):
+                counts[other] += c
+        white = ({model.item_map.get(i) for i in q.white_list}
+                 if q.white_list is not None else None)
+        black = {model.item_map.get(i) for i in (q.black_list or [])}
+        cats = set(q.categories) if q.categories is not None else None
+
+        def ok(i: int) -> bool:
+            if i in qidx or i in black:
+                return False
+            if white is not None and i not in white:
+                return False
+            if cats is not None:
+                meta = model.items.get(model.item_inv[i])
+                if not (meta and meta.categories
+                        and cats & set(meta.categories)):
+                    return False
+            return True
+
+        ranked = sorted(((c, i) for i, c in counts.items() if ok(i)),
+                        reverse=True)[:q.num]
+        return PredictedResult([
+            ItemScore(model.item_inv[i], float(c)) for c, i in ranked])
+
+
+class Serving(BaseServing):
+    def serve(self, query, predictions) -> PredictedResult:
+        return predictions[0]
+
+
+class SimilarProductEngine(EngineFactory):
+    @classmethod
+    def apply(cls) -> Engine:
+        return Engine(
+            data_source_class=DataSource,
+            preparator_class=Preparator,
+            algorithm_class={"als": ALSAlgorithm,
+                             "cooccurrence": CooccurrenceAlgorithm,
+                             "": ALSAlgorithm},
+            serving_class=Serving)

@@ -57,7 +57,14 @@ def run_train(engine: Engine,
               skip_sanity_check: bool = False,
               verbose: bool = False) -> str:
     """Train and persist; returns the engine-instance id
-    (CoreWorkflow.runTrain, CoreWorkflow.scala:45-102)."""
+    (CoreWorkflow.runTrain, CoreWorkflow.scala:45-102).
+
+    Under torchrun (multi-GPU DP over RCCL) every rank trains — the
+    collectives require all ranks — but only rank 0 touches storage."""
+    from predictionio_amd.parallel import dist as pdist
+    if pdist.get_rank() != 0:
+        engine.train(engine_params, skip_sanity_check=skip_sanity_check)
+        return ""
     instances = storage.get_meta_data_engine_instances()
     instance = EngineInstance(
         id="", status="INIT", start_time=utcnow(), end_time=utcnow(),
