@@ -332,7 +332,9 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
       const float dinv = rsqrtf(lkk);
       const float ljk = lane > k ? acc[k] * dinv
                                  : (lane == k ? lkk * dinv : 0.f);
-      Lc[wave][k][lane] = ljk;  // persist column k: Lc[k][j] = L[j][k]
+      // persist column k: Lc[k][j] = L[j][k]. Guard: for F < 64 lanes
+      // >= F would write past the (F+1)-wide row into the next column.
+      if (lane < F) Lc[wave][k][lane] = ljk;
       wave_sync();
 #pragma unroll
       for (int m = k + 1; m < F; ++m) {

@@ -8,37 +8,40 @@
 //  - ecommerce predictKnownUser / predictSimilar
 //    (examples/.../ECommAlgorithm.scala:471-506, 541-599)
 //
-// Design (MI355X-first): scores = Xq . Y^T fused with masking and an
-// on-device per-slice top-K reduction, never materializing the B x N score
-// matrix (N up to 10^7). Grid = (item_slices, user_blocks); each 256-thread
-// workgroup owns a 64-user x item-slice tile:
-//  - 64-item chunks of Y staged through LDS (coalesced), then TRANSPOSED
-//    into per-lane registers (lane l holds item l's full factor row) —
-//    a v1 of this kernel kept 8 users per workgroup and re-read Y B/8
-//    times from HBM (1.3 TB per 4096-query batch, measured 656 ms); with
-//    64 users per block Y traffic drops 8x and the dot loop is
-//    1 fma + 1 LDS-broadcast per score.
-//  - each wave scores 16 users against the 64 staged items: the k-loop
-//    reads xs[u][k] (same address across lanes → LDS broadcast) against
-//    yreg[k] (literal register index).
-//  - per-user running top-K in LDS with a wave-ballot insertion filter:
-//    lanes beating the user's current K-th best serialize through lane 0,
-//    which also applies the per-user banned-list (binary search, short
-//    sorted lists) before insertion.
-// Phase 2 (merging the per-slice candidates) is a small torch.topk on
-// [B, slices*K] host-side tensors.
+// Design (v3, MI355X-first): scores = Xq . Y^T fused with masking and an
+// on-device top-K reduction, never materializing the B x N score matrix
+// (N up to 10^7). Grid = (item_slices, user_blocks of 64).
+//
+//  - LANE = USER: each lane carries its user's query vector in packed
+//    float2 registers (loaded once per workgroup) and a scalar running
+//    K-th-best threshold. Insertions go to the lane's own list — fully
+//    parallel, no cross-lane serialization. (v1 had 8 users/WG and
+//    re-read Y B/8 times from HBM — 1.3 TB per 4096-query batch; v2
+//    tiled 64 users but kept thresholds in a register array indexed by a
+//    16-iteration user loop, forcing a 16x unroll that blew L1I.)
+//  - 64-item chunks of Y staged through LDS (coalesced); each wave scores
+//    its own 16-item quarter: per item the k-loop is an LDS-broadcast
+//    float2 read against the xreg registers — v_pk_fma_f32 packed math,
+//    two independent accumulator chains.
+//  - per-(wave, user) top-K lists in LDS; the 4 waves are independent
+//    candidate groups, so the output carries n_slices*4 groups of K and
+//    the host-side merge (one small torch.topk) folds them.
 //
 // Masks:
-//  - item_mask: optional uint8[N], 1 = globally banned (e.g. unavailable
-//    items, category filter precomputed on device)
-//  - per-user banned list (seen/blacklisted items): CSR int32, binary
-//    search per candidate insertion (lists are short; L1-resident)
+//  - item_mask: optional uint8[N], 1 = globally banned (uniform per item,
+//    checked once per wave)
+//  - per-user banned list (seen/blacklisted items): CSR int32; the owning
+//    lane binary-searches its own list per candidate insertion.
 
 #include <float.h>
 #include <hip/hip_runtime.h>
 
-#define TK_CHUNK 64   // items staged per LDS pass (= wave width)
+#define TK_CHUNK 64   // items staged per LDS pass
 #define TK_MAXK 64    // max supported K
+#define TK_WAVES 4
+#define TK_UPB 64     // users per block (= lane count)
+
+typedef __attribute__((ext_vector_type(2))) float f32x2;
 
 __device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
   int lo = 0, hi = n - 1;
@@ -58,19 +61,16 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     const uint8_t* __restrict__ item_mask,       // N or nullptr
     const long long* __restrict__ ban_indptr,    // B+1 or nullptr
     const int* __restrict__ ban_indices,         // sorted per user
-    float* __restrict__ out_val,         // B x n_slices x K
-    int* __restrict__ out_idx,           // B x n_slices x K
+    float* __restrict__ out_val,         // B x (n_slices*TK_WAVES) x K
+    int* __restrict__ out_idx,
     int B, long long N, int K, int n_slices, int item_base)
 {
-  constexpr int UPW = 16;           // users per wave
-  constexpr int UPB = 4 * UPW;      // users per block = 64
-  // dynamic LDS layout: ys[64][F+1] | xs[UPB][F+1] | topv[UPB][K] |
-  // topi[UPB][K]
+  constexpr int FP = F + 2;  // row stride in floats (float2-aligned, odd/2)
+  // dynamic LDS: ys[TK_CHUNK][FP] | topv[TK_WAVES*TK_UPB][K] | topi[...]
   extern __shared__ float lds[];
-  float* ys = lds;                              // 64 x (F+1)
-  float* xs = ys + TK_CHUNK * (F + 1);          // UPB x (F+1)
-  float* topv = xs + UPB * (F + 1);             // UPB x K
-  int* topi = reinterpret_cast<int*>(topv + UPB * K);
+  float* ys = lds;
+  float* topv = ys + TK_CHUNK * FP;
+  int* topi = reinterpret_cast<int*>(topv + TK_WAVES * TK_UPB * K);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -78,31 +78,50 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
 
   const int slice = blockIdx.x;
   const int ublock = blockIdx.y;
-  const long long u0 = (long long)ublock * UPB;
+  const long long u0 = (long long)ublock * TK_UPB;
+  const long long guser = u0 + lane;
+  const bool has_user = guser < B;
 
   // item range of this slice
   const long long per = (N + n_slices - 1) / n_slices;
   const long long it0 = (long long)slice * per;
   const long long it1 = min(N, it0 + per);
 
-  // stage user query vectors + init top-K state
-  for (int e = tid; e < UPB * F; e += 256) {
+  // ---- load query vectors: coalesced stage into ys, transpose to regs
+  for (int e = tid; e < TK_UPB * F; e += 256) {
     const int u = e / F;
     const int k = e % F;
-    xs[u * (F + 1) + k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
+    ys[u * FP + k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
   }
-  for (int e = tid; e < UPB * K; e += 256) {
+  __syncthreads();
+  f32x2 xreg[F / 2];
+  {
+    const float* xrow = ys + lane * FP;
+#pragma unroll
+    for (int q = 0; q < F / 2; ++q)
+      xreg[q] = f32x2{xrow[2 * q], xrow[2 * q + 1]};
+  }
+  __syncthreads();
+
+  // ---- init this block's top-K lists
+  for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
   __syncthreads();
 
-  // running K-th-best threshold per user, register-replicated across the
-  // wave (intra-wave LDS cross-lane communication is not ordered without a
-  // barrier — shuffles are)
-  float uth[UPW];
-#pragma unroll
-  for (int uu = 0; uu < UPW; ++uu) uth[uu] = -FLT_MAX;
+  float th = -FLT_MAX;               // this lane's K-th best (this wave)
+  float* tvu = topv + (wave * TK_UPB + lane) * K;
+  int* tiu = topi + (wave * TK_UPB + lane) * K;
+
+  // this lane's banned list
+  const int* ban = nullptr;
+  int bn = 0;
+  if (ban_indptr != nullptr && has_user) {
+    const long long b0 = ban_indptr[guser];
+    bn = (int)(ban_indptr[guser + 1] - b0);
+    ban = ban_indices + b0;
+  }
 
   for (long long base = it0; base < it1; base += TK_CHUNK) {
     const int cn = (int)min((long long)TK_CHUNK, it1 - base);
@@ -111,86 +130,55 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     for (int e = tid; e < cn * F; e += 256) {
       const int c = e / F;
       const int k = e % F;
-      ys[c * (F + 1) + k] = Y[(base + c) * F + k];
+      ys[c * FP + k] = Y[(base + c) * F + k];
     }
     __syncthreads();
 
-    // transpose the chunk into registers: lane l = item l of the chunk
-    // (stride F+1 across lanes → conflict-free column reads)
-    float yreg[F];
-    const bool live = lane < cn;
-    {
-      const float* yrow = ys + lane * (F + 1);
+    // wave w scores items [16w, 16w+16) of the chunk for all 64 users
+    const int c_lo = wave * (TK_CHUNK / TK_WAVES);
+    const int c_hi = min(cn, c_lo + TK_CHUNK / TK_WAVES);
+    for (int c = c_lo; c < c_hi; ++c) {
+      const long long item = base + c;
+      if (item_mask != nullptr && item_mask[item]) continue;
+      const f32x2* yrow = reinterpret_cast<const f32x2*>(ys + c * FP);
+      f32x2 a0 = {0.f, 0.f}, a1 = {0.f, 0.f};
 #pragma unroll
-      for (int k = 0; k < F; ++k) yreg[k] = live ? yrow[k] : 0.f;
-    }
-    const long long item = base + lane;
-    const bool open = live &&
-        !(item_mask != nullptr && item_mask[item]);
-
-    // each wave scores its UPW users against the 64 staged items
-#pragma unroll
-    for (int uu = 0; uu < UPW; ++uu) {
-      const int u = wave * UPW + uu;
-      const long long guser = u0 + u;
-      float s = -FLT_MAX;
-      if (open && guser < B) {
-        const float* xrow = xs + u * (F + 1);
-        float acc = 0.f;
-#pragma unroll
-        for (int k = 0; k < F; ++k) acc = fmaf(xrow[k], yreg[k], acc);
-        s = acc;
+      for (int q = 0; q + 1 < F / 2; q += 2) {
+        a0 += xreg[q] * yrow[q];       // v_pk_fma_f32, 2 indep chains
+        a1 += xreg[q + 1] * yrow[q + 1];
       }
-      // wave-ballot insertion: only lanes beating the running threshold
-      unsigned long long mask = __ballot(s > uth[uu]);
-      while (mask) {
-        const int src = __ffsll(mask) - 1;
-        mask &= mask - 1;
-        const float v = __shfl(s, src);
-        const long long cand = base + src;
-        float nth = uth[uu];
-        if (lane == 0) {
-          // optional per-user banned-list check (short sorted list);
-          // topv/topi rows of this user are touched by lane 0 only inside
-          // the scan loop, so no cross-lane LDS hazard here
-          bool banned = false;
-          if (ban_indptr != nullptr) {
-            const long long bs0 = ban_indptr[guser];
-            const int bn = (int)(ban_indptr[guser + 1] - bs0);
-            banned = in_sorted(ban_indices + bs0, bn, (int)(cand + item_base));
-          }
-          if (!banned && v > nth) {
-            // replace current min of the K-list
-            float* tvu = topv + u * K;
-            int* tiu = topi + u * K;
-            int mi = 0;
-            float mv = tvu[0];
-            for (int q = 1; q < K; ++q)
-              if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
-            tvu[mi] = v;
-            tiu[mi] = (int)(cand + item_base);
-            // new threshold = K-th best = new min of the list
-            float nm = tvu[0];
-            for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
-            nth = nm;
-          }
+      if (F / 2 & 1) a0 += xreg[F / 2 - 1] * yrow[F / 2 - 1];
+      const float s = a0.x + a0.y + a1.x + a1.y;
+      if (has_user && s > th) {
+        if (ban == nullptr ||
+            !in_sorted(ban, bn, (int)(item + item_base))) {
+          // replace current min of this lane's K-list
+          int mi = 0;
+          float mv = tvu[0];
+          for (int q = 1; q < K; ++q)
+            if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
+          tvu[mi] = s;
+          tiu[mi] = (int)(item + item_base);
+          float nm = tvu[0];
+          for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
+          th = nm;
         }
-        uth[uu] = __shfl(nth, 0);
-        if (mask) mask &= __ballot(s > uth[uu]);
       }
     }
   }
   __syncthreads();
 
-  // write out per-slice candidates
-  for (int e = tid; e < UPB * K; e += 256) {
-    const int u = e / K;
+  // ---- write out: group g = slice*TK_WAVES + wave, K entries per user
+  for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
+    const int w = e / (TK_UPB * K);
+    const int u = (e / K) % TK_UPB;
     const int q = e % K;
-    const long long guser = u0 + u;
-    if (guser < B) {
-      const long long o = (guser * n_slices + slice) * K + q;
-      out_val[o] = topv[u * K + q];
-      out_idx[o] = topi[u * K + q];
+    const long long gu = u0 + u;
+    if (gu < B) {
+      const long long g = (long long)slice * TK_WAVES + w;
+      const long long o = (gu * n_slices * TK_WAVES + g) * K + q;
+      out_val[o] = topv[(w * TK_UPB + u) * K + q];
+      out_idx[o] = topi[(w * TK_UPB + u) * K + q];
     }
   }
 }
@@ -202,14 +190,13 @@ extern "C" void launch_topk_score(
     int B, long long N, int f, int K, int n_slices, int item_base,
     hipStream_t stream)
 {
-  const int UPB = 64;
-  dim3 grid(n_slices, (B + UPB - 1) / UPB);
+  dim3 grid(n_slices, (B + TK_UPB - 1) / TK_UPB);
   dim3 block(256);
 #define LAUNCH(FF)                                                          \
   do {                                                                      \
     const size_t lds_bytes =                                                \
-        sizeof(float) * ((TK_CHUNK + UPB) * (FF + 1) + UPB * K) +           \
-        sizeof(int) * UPB * K;                                              \
+        sizeof(float) * (TK_CHUNK * (FF + 2)) +                             \
+        (sizeof(float) + sizeof(int)) * TK_WAVES * TK_UPB * K;              \
     static bool attr_set_##FF = false;                                      \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
       hipFuncSetAttribute(                                                  \

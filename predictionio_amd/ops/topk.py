@@ -34,9 +34,12 @@ def topk_score(Xq: torch.Tensor, Y: torch.Tensor, K: int,
         Xp = Xq if pf == f else torch.nn.functional.pad(Xq, (0, pf - f))
         Yp = Y if pf == f else torch.nn.functional.pad(Y, (0, pf - f))
         if n_slices is None:
-            # enough (slice, ublock) workgroups to fill 256 CUs; slices add
-            # merge cost linearly, so scale with N
-            n_slices = max(1, min(1024, N // 4096))
+            # enough (slice, ublock) workgroups to oversubscribe 256 CUs
+            # (~2048 WGs); more slices only add insert + merge cost, so
+            # scale them inversely with the user-block count
+            ublocks = (B + 63) // 64
+            n_slices = max(1, min(2048 // ublocks + 1,
+                                  (N + 255) // 256))
         vals, idxs = hip_ext().topk_score(
             Xp.contiguous(), Yp.contiguous(), K, int(n_slices),
             item_mask.contiguous() if item_mask is not None else None,
