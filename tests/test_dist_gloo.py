@@ -117,3 +117,62 @@ class TestDistributed:
             t.step()
         X1 = torch.tensor(res[0][0])
         assert torch.allclose(X1, t.X, atol=1e-4, rtol=1e-4)
+
+
+def _exchange(rank, world):
+    from predictionio_amd.parallel import dist as pdist
+    g = torch.Generator().manual_seed(7 + rank)
+    n_cols = 10
+    # each rank holds random triples; after exchange each rank must hold
+    # exactly the triples (from all ranks) whose col is in its block
+    rows = torch.randint(0, 100, (20,), generator=g, dtype=torch.int32)
+    cols = torch.randint(0, n_cols, (20,), generator=g, dtype=torch.int32)
+    vals = torch.rand(20, generator=g)
+    r2, c2, v2 = pdist.exchange_triples(rows, cols, vals, n_cols)
+    lo, hi = pdist.block_bounds(n_cols, world, rank)
+    assert ((c2 >= lo) & (c2 < hi)).all(), "received out-of-block col"
+    mine = sorted(zip(r2.tolist(), c2.tolist(),
+                      [round(v, 5) for v in v2.tolist()]))
+    return mine
+
+
+def _bench_shard_path(rank, world):
+    """Exercise bench.py's synth_shard + trainer sharded setup on gloo."""
+    import bench
+    from predictionio_amd.models.als import ALSParams, ALSTrainer
+    device = torch.device("cpu")
+    p = ALSParams(rank=16, iterations=1, lambda_=0.01, alpha=10.0,
+                  implicit=True, seed=5)
+    trainer = ALSTrainer(p, n_users=8 * world, n_items=12, device=device)
+    (u, i, v), (ii, iu, iv) = bench.synth_shard(trainer, nnz_per_user=4,
+                                                seed=5, device=device)
+    trainer.set_ratings_sharded((u, i, v), (ii - trainer.i_lo, iu, iv))
+    trainer.init_factors()
+    trainer.step()
+    X, Y = trainer.gather_factors()
+    assert torch.isfinite(X).all() and torch.isfinite(Y).all()
+    return [X.shape[0], Y.shape[0]]
+
+
+class TestExchangeTriples:
+    def test_partition_by_col(self):
+        res = _spawn("_exchange", port=29617)
+        # every triple generated must appear exactly once across ranks
+        all_triples = sorted(res[0] + res[1])
+        regen = []
+        for rank in range(2):
+            g = torch.Generator().manual_seed(7 + rank)
+            rows = torch.randint(0, 100, (20,), generator=g,
+                                 dtype=torch.int32)
+            cols = torch.randint(0, 10, (20,), generator=g,
+                                 dtype=torch.int32)
+            vals = torch.rand(20, generator=g)
+            regen += list(zip(rows.tolist(), cols.tolist(),
+                              [round(v, 5) for v in vals.tolist()]))
+        assert all_triples == sorted(regen)
+
+
+class TestBenchDistPath:
+    def test_bench_shard_setup(self):
+        res = _spawn("_bench_shard_path", port=29619)
+        assert res[0] == [16, 12] and res[1] == [16, 12]
