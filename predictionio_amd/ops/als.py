@@ -26,33 +26,38 @@ def gramian(Y: torch.Tensor) -> torch.Tensor:
     return Y.t() @ Y
 
 
-def woodbury_w(Y: torch.Tensor, YtY: torch.Tensor,
-               lam: float) -> torch.Tensor:
-    """W = (YtY + lam I)^-1 Y — the per-half-iteration precompute that
-    enables the per-row Woodbury fast path (als_woodbury_kernel). One
-    64x64 Cholesky + one triangular-solve GEMM over all items (rocBLAS)."""
+def woodbury_lv(Y: torch.Tensor, YtY: torch.Tensor,
+                lam: float) -> tuple:
+    """(L, V): L = chol(YtY + lam I), V = Y L^-T — the per-half-iteration
+    precompute that enables the per-row Woodbury fast path
+    (als_woodbury_kernel). One fxf Cholesky + one triangular-solve GEMM
+    over all items (rocBLAS trsm); the whitened rows satisfy
+    v_i . v_j = y_i^T B^-1 y_j."""
     f = Y.shape[1]
     B = YtY + lam * torch.eye(f, dtype=Y.dtype, device=Y.device)
     L = torch.linalg.cholesky(B)
-    return torch.cholesky_solve(Y.t().contiguous(), L).t().contiguous()
+    # V L^T = Y  →  V = Y L^-T  (row v_i = L^-1 y_i)
+    V = torch.linalg.solve_triangular(L.mT, Y, upper=True,
+                                      left=False).contiguous()
+    return L, V
 
 
 def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
               values: torch.Tensor, Y: torch.Tensor,
               YtY: Optional[torch.Tensor] = None,
               lam: float = 0.01, alpha: float = 1.0,
-              implicit: bool = False, wr_scale: bool = True,
-              W: Optional[torch.Tensor] = None) -> torch.Tensor:
+              implicit: bool = False, wr_scale: bool = True) -> torch.Tensor:
     """Solve all rows of one ALS half-iteration.
 
     explicit: (sum y y^T + lam*nnz*I) x = sum r*y      (ALS-WR, like MLlib)
     implicit: (YtY + sum alpha*r y y^T + lam*I) x = sum (1+alpha*r) y
               (Hu-Koren; YtY required)
 
-    On GPU, rows with nnz <= 32 take the Woodbury path (exact — same
-    linear system solved through the push-through identity); pass a
-    precomputed W = woodbury_w(Y, YtY, lam) to amortize it across calls,
-    else it is computed here for implicit mode.
+    On GPU, rows with nnz <= 32 take the Woodbury path (exact — the same
+    linear system through the push-through identity). Implicit mode runs
+    three phases: the Woodbury kernel emits whitened Z rows, one rocBLAS
+    triangular solve maps Z -> X for all rows, then the dense kernel fills
+    the big (nnz > 32) rows directly.
     """
     if implicit:
         wr_scale = False  # Hu-Koren regularizes with plain lambda*I
@@ -60,23 +65,32 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
             YtY = gramian(Y)
     if Y.is_cuda:
         from predictionio_amd.ops import hip_ext
+        ext = hip_ext()
         f = Y.shape[1]
         pf = pad_rank(f)
-        Yp = Y if pf == f else torch.nn.functional.pad(Y, (0, pf - f))
+        Yp = (Y if pf == f
+              else torch.nn.functional.pad(Y, (0, pf - f))).contiguous()
         YtYp = None
         if YtY is not None:
             YtYp = (YtY if pf == f
-                    else torch.nn.functional.pad(YtY, (0, pf - f, 0, pf - f)))
-        if implicit and W is None and pf <= 64:
-            W = woodbury_w(Yp, YtYp, lam)
-        elif W is not None and pf != f:
-            W = torch.nn.functional.pad(W, (0, pf - f))
-        X = hip_ext().als_solve(
-            indptr.contiguous(), indices.contiguous(), values.contiguous(),
-            Yp.contiguous(),
-            YtYp.contiguous() if YtYp is not None else None,
-            W.contiguous() if W is not None else None,
-            float(lam), float(alpha), bool(implicit), bool(wr_scale))
+                    else torch.nn.functional.pad(YtY, (0, pf - f, 0, pf - f))
+                    ).contiguous()
+        ip, ix, vv = (indptr.contiguous(), indices.contiguous(),
+                      values.contiguous())
+        if implicit and pf <= 64:
+            L, V = woodbury_lv(Yp, YtYp, lam)
+            Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
+                              float(alpha), True, False, 1, None)
+            # X = Z L^-1 for all rows (Woodbury rows hold z; big rows get
+            # overwritten by the dense pass next)
+            X = torch.linalg.solve_triangular(L, Z, upper=False, left=False)
+            X = X.contiguous()
+            ext.als_solve(ip, ix, vv, Yp, YtYp, None, float(lam),
+                          float(alpha), True, False, 2, X)
+        else:
+            X = ext.als_solve(ip, ix, vv, Yp, YtYp, None, float(lam),
+                              float(alpha), bool(implicit), bool(wr_scale),
+                              0, None)
         return X[:, :f].contiguous() if pf != f else X
     return als_solve_ref(indptr, indices, values, Y, YtY, lam, alpha,
                          implicit, wr_scale)

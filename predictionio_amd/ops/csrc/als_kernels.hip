@@ -369,34 +369,41 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
 //
 // Instead of building + factorizing the FxF normal matrix per row, exploit
 // that only nnz rank-1 terms differ from a FIXED base B:
-//   implicit (Hu-Koren):  A_u = B + U C U^T,  B = Y^T Y + lambda I,
+//   implicit (Hu-Koren):  A_u = B + U C U^T,  B = Y^T Y + lambda I = L L^T,
 //     U = [y_i] (F x n), C = diag(alpha r_i), b_u = U c2, c2_i = 1+alpha r_i.
-//     Push-through identity:  x_u = W_u (I + C G)^-1 c2,
-//     where W = B^-1 Y is precomputed ONCE per half-iteration (host GEMM)
-//     and G = U^T B^-1 U = Y_u . W_u^T (n x n, symmetric).
-//     Symmetrized with D = C^(1/2):  solve (I + D G D) t = c2 / d, s = d t,
-//     x = sum_i s_i w_i.   (I + DGD is SPD.)
+//     Push-through identity:  x_u = B^-1 U (I + C G)^-1 c2,
+//     with G = U^T B^-1 U.  Using the WHITENED factors V = Y L^-T
+//     (precomputed once per half-iteration — one host-side trsm):
+//       G = V_u . V_u^T   (v_i = L^-1 y_i, so v_i.v_j = y_i^T B^-1 y_j)
+//     Symmetrized with D = C^(1/2): solve (I + D G D) t = c2 / d, s = d t,
+//     and emit z_u = sum_i s_i v_i; the host maps X = Z L^-1 for ALL rows
+//     in one triangular-solve GEMM (x = L^-T z).  Only ONE staged factor
+//     matrix per row — half the HBM traffic and LDS of the W = B^-1 Y
+//     formulation, doubling resident waves.
 //   explicit (ALS-WR):  A_u = U U^T + reg I, b_u = U r:
-//     x_u = U (G + reg I)^-1 r with G = Y_u . Y_u^T  (W never needed).
+//     x_u = U (G + reg I)^-1 r with G = Y_u . Y_u^T — same kernel body
+//     with Y staged instead of V and X emitted directly (no host solve).
 //
-// Cost per row: n^2 F MACs for G + n^3/3 solve — ~5x fewer FLOPs than the
-// FxF Cholesky at n=20, F=64, with NO long serial dependency chains.
+// Cost per row: n^2/2 F MACs for G + n^3/3 solve — ~5x fewer FLOPs than
+// the FxF Cholesky at n=20, F=64, with NO long serial dependency chains.
 // One wave per row, 2 waves per workgroup; rows with nnz > WOODBURY_MAX_NNZ
-// are skipped here and handled by the workgroup kernel (skip_small=1).
+// are skipped here and handled by the wave/workgroup kernels.
 // LDS rows padded to F+4 floats so G-dot reads from row i (stride 68) land
 // in different bank groups per lane.
 // ---------------------------------------------------------------------------
 
 #define WOODBURY_MAX_NNZ 32
 
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
 template <int F>
 __global__ __launch_bounds__(128) void als_woodbury_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
     const float* __restrict__ values,
-    const float* __restrict__ Y,
-    const float* __restrict__ W,     // B^-1 Y (implicit mode), else nullptr
-    float* __restrict__ X,
+    const float* __restrict__ Y,     // item factors (explicit mode)
+    const float* __restrict__ V,     // whitened factors Y L^-T (implicit)
+    float* __restrict__ X,           // X (explicit) / Z (implicit)
     int n_rows,
     float lambda,
     float alpha,
@@ -408,20 +415,19 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  __shared__ float Yl[2][NW][FP];
-  __shared__ float Wl[2][NW][FP];
+  __shared__ float Yl[2][NW][FP];      // staged factor rows (Y or V)
   __shared__ float M[2][NW][NW + 1];   // I+DGD (implicit) / G+regI (explicit)
   __shared__ float tv[2][NW];          // rhs, then solution t
   __shared__ float dv[2][NW];          // D diagonal (implicit)
 
   float* yl = &Yl[wave][0][0];
-  float* wl = implicit_mode ? &Wl[wave][0][0] : &Yl[wave][0][0];
+  const float* src = implicit_mode ? V : Y;
 
   for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
        row += (long long)gridDim.x * 2) {
     const long long start = indptr[row];
     const int n = (int)(indptr[row + 1] - start);
-    if (n > NW) continue;  // workgroup kernel handles these
+    if (n > NW) continue;  // wave/workgroup kernel handles these
     if (n == 0) {
       if (lane < F) X[row * (long long)F + lane] = 0.f;
       continue;
@@ -430,10 +436,7 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     // ---- stage factor rows + per-item weights ----
     for (int c = 0; c < n; ++c) {
       const long long col = indices[start + c];
-      if (lane < F) {
-        yl[c * FP + lane] = Y[col * F + lane];
-        if (implicit_mode) Wl[wave][c][lane] = W[col * F + lane];
-      }
+      if (lane < F) yl[c * FP + lane] = src[col * F + lane];
     }
     if (lane < n) {
       const float r = values[start + lane];
@@ -462,18 +465,12 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
       while (p >= n - i && i < n) { p -= n - i; ++i; }
       int j = i + p;
       for (int pp = lane; pp < npairs; pp += 64) {
-        float dot = 0.f;
-        const float* yi = &yl[i * FP];
-        const float* wj = &wl[j * FP];
+        const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
+        const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
+        f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};  // 2x v_pk_fma_f32 per q
 #pragma unroll
-        for (int q = 0; q < F / 4; ++q) {
-          const float4 a4 = *reinterpret_cast<const float4*>(yi + 4 * q);
-          const float4 b4 = *reinterpret_cast<const float4*>(wj + 4 * q);
-          dot = fmaf(a4.x, b4.x, dot);
-          dot = fmaf(a4.y, b4.y, dot);
-          dot = fmaf(a4.z, b4.z, dot);
-          dot = fmaf(a4.w, b4.w, dot);
-        }
+        for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
+        float dot = acc4.x + acc4.y + acc4.z + acc4.w;
         if (implicit_mode) {
           dot *= dv[wave][i] * dv[wave][j];
           if (i == j) dot += 1.f;
@@ -524,11 +521,12 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     if (implicit_mode && lane < n) tv[wave][lane] *= dv[wave][lane];  // s = d t
     wave_sync();
 
-    // ---- x = sum_i s_i w_i  (implicit)  or  sum_i s_i y_i  (explicit) ----
+    // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
+    //      or   sum_i s_i y_i (explicit: x directly) ----
     if (lane < F) {
       float x = 0.f;
       for (int c = 0; c < n; ++c)
-        x = fmaf(tv[wave][c], wl[c * FP + lane], x);
+        x = fmaf(tv[wave][c], yl[c * FP + lane], x);
       X[row * (long long)F + lane] = x;
     }
     wave_sync();
@@ -543,33 +541,39 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
 
 extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
-    const float* Y, const float* YtY, const float* W, float* X,
+    const float* Y, const float* YtY, const float* V, float* X,
     int n_rows, int f, float lambda, float alpha,
-    int implicit_mode, int wr_scale, hipStream_t stream)
+    int implicit_mode, int wr_scale, int which, hipStream_t stream)
 {
   // >> 256 workgroups to fill 256 CUs across 8 XCDs; grid-stride for huge
-  // row counts. W != nullptr (implicit) or explicit mode with f <= 64
-  // enables the Woodbury fast path for rows with nnz <= WOODBURY_MAX_NNZ;
-  // the dense-Gramian kernels then skip those rows.
+  // row counts.
+  // which: 0 = both passes (explicit mode — X is direct either way),
+  //        1 = Woodbury only (implicit: emits Z for the host-side trsm),
+  //        2 = dense only, skipping Woodbury-owned rows (implicit: fills
+  //            the big rows with direct X after the trsm).
   if (n_rows <= 0) return;
   // wave/woodbury kernels: 2 rows per 128-thread workgroup
   long long wg = ((long long)n_rows + 1) / 2;
   int grid_w = (int)(wg < (1 << 20) ? wg : (1 << 20));
   int grid_b = n_rows < (1 << 20) ? n_rows : (1 << 20);
-  const bool woodbury = f <= 64 && (implicit_mode ? W != nullptr : true);
-  const int skip = woodbury ? WOODBURY_MAX_NNZ : -1;
+  const bool woodbury = f <= 64 && which != 2 &&
+      (implicit_mode ? V != nullptr : true);
+  const int skip = (woodbury || which == 2) ? WOODBURY_MAX_NNZ : -1;
+  const bool dense = which != 1;
 #define LAUNCH_WAVE(FF)                                                      \
+  if (dense)                                                                 \
   hipLaunchKernelGGL((als_solve_wave_kernel<FF>), dim3(grid_w), dim3(128),   \
                      0, stream, indptr, indices, values, Y, YtY, X, n_rows,  \
                      lambda, alpha, implicit_mode, wr_scale, skip)
 #define LAUNCH_BLOCK(FF)                                                     \
+  if (dense)                                                                 \
   hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,     \
                      stream, indptr, indices, values, Y, YtY, X, n_rows,     \
                      lambda, alpha, implicit_mode, wr_scale, skip)
 #define LAUNCH_WOODBURY(FF)                                                  \
   if (woodbury)                                                              \
   hipLaunchKernelGGL((als_woodbury_kernel<FF>), dim3(grid_w), dim3(128),     \
-                     0, stream, indptr, indices, values, Y, W, X, n_rows,    \
+                     0, stream, indptr, indices, values, Y, V, X, n_rows,    \
                      lambda, alpha, implicit_mode, wr_scale)
   switch (f) {
     case 16: LAUNCH_WOODBURY(16); LAUNCH_WAVE(16); break;

@@ -12,9 +12,9 @@
 
 extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
-    const float* Y, const float* YtY, const float* W, float* X,
+    const float* Y, const float* YtY, const float* V, float* X,
     int n_rows, int f, float lambda, float alpha,
-    int implicit_mode, int wr_scale, hipStream_t stream);
+    int implicit_mode, int wr_scale, int which, hipStream_t stream);
 
 extern "C" void launch_topk_score(
     const float* Xq, const float* Y, const uint8_t* item_mask,
@@ -39,15 +39,17 @@ bool supported_rank(int64_t f) {
 //   explicit: (sum_i y_i y_i^T + lambda*nnz_r*I) x_r = sum_i r_i y_i
 //   implicit: (YtY + sum_i alpha*r * y_i y_i^T + lambda*I) x_r
 //               = sum_i (1+alpha*r) y_i
-// W = (YtY + lambda I)^-1 Y precomputed on the host side per half-iteration
-// (torch cholesky_solve); enables the per-row Woodbury fast path for small
-// nnz in implicit mode (see als_woodbury_kernel).
+// V = Y L^-T (L = chol(YtY + lambda I)) precomputed on the host per
+// half-iteration enables the per-row Woodbury fast path in implicit mode
+// (see als_woodbury_kernel). which: 0 = both passes, 1 = Woodbury rows
+// only (emits Z), 2 = dense rows only (writes into `out`).
 torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
                         torch::Tensor values, torch::Tensor Y,
                         c10::optional<torch::Tensor> YtY,
-                        c10::optional<torch::Tensor> W,
+                        c10::optional<torch::Tensor> V,
                         double lambda, double alpha,
-                        bool implicit_mode, bool wr_scale) {
+                        bool implicit_mode, bool wr_scale,
+                        int64_t which, c10::optional<torch::Tensor> out) {
   TORCH_CHECK(indptr.is_cuda() && indptr.scalar_type() == torch::kInt64 &&
                   indptr.is_contiguous(), "indptr must be contiguous i64 GPU");
   TORCH_CHECK(indices.is_cuda() && indices.scalar_type() == torch::kInt32 &&
@@ -64,20 +66,28 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
     TORCH_CHECK(YtY->size(0) == f && YtY->size(1) == f, "YtY must be f x f");
     yty_ptr = YtY->data_ptr<float>();
   }
-  const float* w_ptr = nullptr;
-  if (W.has_value()) {
-    check_cuda_f32(*W, "W");
-    TORCH_CHECK(W->sizes() == Y.sizes(), "W must match Y shape");
-    w_ptr = W->data_ptr<float>();
+  const float* v_ptr = nullptr;
+  if (V.has_value()) {
+    check_cuda_f32(*V, "V");
+    TORCH_CHECK(V->sizes() == Y.sizes(), "V must match Y shape");
+    v_ptr = V->data_ptr<float>();
   }
-  auto X = torch::empty({n_rows, f}, Y.options());
+  torch::Tensor X;
+  if (out.has_value()) {
+    check_cuda_f32(*out, "out");
+    TORCH_CHECK(out->size(0) == n_rows && out->size(1) == f,
+                "out must be n_rows x f");
+    X = *out;
+  } else {
+    X = torch::empty({n_rows, f}, Y.options());
+  }
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   launch_als_solve(reinterpret_cast<const long long*>(indptr.data_ptr<int64_t>()), indices.data_ptr<int>(),
                    values.data_ptr<float>(), Y.data_ptr<float>(), yty_ptr,
-                   w_ptr, X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
+                   v_ptr, X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
                    (float)alpha, implicit_mode ? 1 : 0, wr_scale ? 1 : 0,
-                   stream);
+                   (int)which, stream);
   C10_HIP_CHECK(hipGetLastError());
   return X;
 }
@@ -138,10 +148,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X (gfx950) HIP kernels for predictionio_amd";
   m.def("als_solve", &als_solve, "Fused ALS Gramian+Cholesky half-iteration",
         py::arg("indptr"), py::arg("indices"), py::arg("values"),
-        py::arg("Y"), py::arg("YtY") = py::none(), py::arg("W") = py::none(),
+        py::arg("Y"), py::arg("YtY") = py::none(), py::arg("V") = py::none(),
         py::arg("lambda_") = 0.01,
         py::arg("alpha") = 1.0, py::arg("implicit_mode") = false,
-        py::arg("wr_scale") = true);
+        py::arg("wr_scale") = true, py::arg("which") = 0,
+        py::arg("out") = py::none());
   m.def("topk_score", &topk_score, "Fused masked top-K scoring",
         py::arg("Xq"), py::arg("Y"), py::arg("K"), py::arg("n_slices") = 64,
         py::arg("item_mask") = py::none(), py::arg("ban_indptr") = py::none(),
