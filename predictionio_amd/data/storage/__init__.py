@@ -91,6 +91,12 @@ def _get_client(source_name: str, cfg: dict):
             from predictionio_amd.data.storage.localfs import LocalFSClient
             client = LocalFSClient(cfg.get(
                 "path", os.path.join(_base_dir(), "models")))
+        elif typ == "fsspec":
+            from predictionio_amd.data.storage.fsspec_store import (
+                FsspecClient,
+            )
+            client = FsspecClient(cfg.get(
+                "path", os.path.join(_base_dir(), "models")))
         else:
             raise StorageError(f"Unknown storage source type: {typ}")
         _sources[key] = client
@@ -124,6 +130,9 @@ def _dao(repo: str, kind: str):
     elif typ == "localfs":
         from predictionio_amd.data.storage import localfs as be
         table = {"models": be.LocalFSModels}
+    elif typ == "fsspec":
+        from predictionio_amd.data.storage import fsspec_store as be
+        table = {"models": be.FsspecModels}
     else:
         raise StorageError(f"Unknown storage type {typ}")
     if kind not in table:

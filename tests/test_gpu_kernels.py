@@ -180,3 +180,43 @@ class TestALSTrainerGPU:
         Xg, Yg = tg.fit()
         assert torch.allclose(Xg.cpu(), Xc, atol=5e-2, rtol=5e-2), \
             f"max diff {(Xg.cpu() - Xc).abs().max()}"
+
+
+@requires_gpu
+class TestTemplatesOnGPU:
+    def test_recommendation_template_gpu(self, mem_storage):
+        """Template end-to-end on the MI355X: train on synthetic events
+        (kernels + storage together), predict through the fused top-K."""
+        import random
+        from predictionio_amd.data.events import DataMap, Event, utcnow
+        from predictionio_amd.data.storage.base import App
+        app_id = mem_storage.get_meta_data_apps().insert(
+            App(id=0, name="GpuApp"))
+        mem_storage.get_l_events().init(app_id)
+        rng = random.Random(5)
+        le = mem_storage.get_l_events()
+        for u in range(50):
+            for i in rng.sample(range(30), 8):
+                le.insert(Event(
+                    event="rate", entity_type="user", entity_id=f"u{u}",
+                    target_entity_type="item", target_entity_id=f"i{i}",
+                    properties=DataMap({"rating": rng.uniform(1, 5)}),
+                    event_time=utcnow()), app_id)
+        from predictionio_amd.controller import EngineParams, Params
+        from predictionio_amd.templates.recommendation import (
+            ALSAlgorithm, Query, RecommendationEngine,
+        )
+        e = RecommendationEngine.apply()
+        ep = EngineParams(
+            data_source_params=Params({"appName": "GpuApp"}),
+            algorithms_params=[("als", Params(
+                {"rank": 16, "numIterations": 5, "seed": 3}))])
+        models = e.train(ep)
+        assert models[0].user_features.is_cuda  # trained on device
+        algo = ALSAlgorithm(ep.algorithms_params[0][1])
+        r = algo.predict(models[0], Query(user="u1", num=5))
+        assert len(r.item_scores) == 5
+        banned = [s.item for s in r.item_scores[:2]]
+        r2 = algo.predict(models[0],
+                          Query(user="u1", num=5, black_list=banned))
+        assert not set(s.item for s in r2.item_scores) & set(banned)
