@@ -77,7 +77,7 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
                     ).contiguous()
         ip, ix, vv = (indptr.contiguous(), indices.contiguous(),
                       values.contiguous())
-        if implicit and pf <= 64:
+        if implicit and pf <= 128:
             L, V = woodbury_lv(Yp, YtYp, lam)
             Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
                               float(alpha), True, False, 1, None)

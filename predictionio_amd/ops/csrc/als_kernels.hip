@@ -429,14 +429,17 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     const int n = (int)(indptr[row + 1] - start);
     if (n > NW) continue;  // wave/workgroup kernel handles these
     if (n == 0) {
-      if (lane < F) X[row * (long long)F + lane] = 0.f;
+      for (int e = lane; e < F; e += 64)
+        X[row * (long long)F + e] = 0.f;
       continue;
     }
 
     // ---- stage factor rows + per-item weights ----
+    // (strided so F = 128 works with 64 lanes)
     for (int c = 0; c < n; ++c) {
       const long long col = indices[start + c];
-      if (lane < F) yl[c * FP + lane] = src[col * F + lane];
+      for (int e = lane; e < F; e += 64)
+        yl[c * FP + e] = src[col * F + e];
     }
     if (lane < n) {
       const float r = values[start + lane];
@@ -523,11 +526,11 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
 
     // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
     //      or   sum_i s_i y_i (explicit: x directly) ----
-    if (lane < F) {
+    for (int e = lane; e < F; e += 64) {
       float x = 0.f;
       for (int c = 0; c < n; ++c)
-        x = fmaf(tv[wave][c], yl[c * FP + lane], x);
-      X[row * (long long)F + lane] = x;
+        x = fmaf(tv[wave][c], yl[c * FP + e], x);
+      X[row * (long long)F + e] = x;
     }
     wave_sync();
   }
@@ -556,7 +559,7 @@ extern "C" void launch_als_solve(
   long long wg = ((long long)n_rows + 1) / 2;
   int grid_w = (int)(wg < (1 << 20) ? wg : (1 << 20));
   int grid_b = n_rows < (1 << 20) ? n_rows : (1 << 20);
-  const bool woodbury = f <= 64 && which != 2 &&
+  const bool woodbury = which != 2 &&
       (implicit_mode ? V != nullptr : true);
   const int skip = (woodbury || which == 2) ? WOODBURY_MAX_NNZ : -1;
   const bool dense = which != 1;
@@ -579,7 +582,7 @@ extern "C" void launch_als_solve(
     case 16: LAUNCH_WOODBURY(16); LAUNCH_WAVE(16); break;
     case 32: LAUNCH_WOODBURY(32); LAUNCH_WAVE(32); break;
     case 64: LAUNCH_WOODBURY(64); LAUNCH_WAVE(64); break;
-    case 128: LAUNCH_BLOCK(128); break;
+    case 128: LAUNCH_WOODBURY(128); LAUNCH_BLOCK(128); break;
     default: break;  // caller validates
   }
 #undef LAUNCH_WAVE
