@@ -248,7 +248,8 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  __shared__ float ys[2][2][F];        // double-buffered staged y per wave
+  constexpr int PD = 4;  // global-load pipeline depth (latency hiding)
+  __shared__ float ys[2][PD][F];       // staged y slots per wave
   __shared__ float Lc[2][F][F + 1];    // persisted L columns: Lc[w][k][j] = L[j][k]
 
   for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
@@ -263,43 +264,53 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
     float b_reg = 0.f;
 
     // ---- Gramian accumulation (register row, LDS broadcast) ----
-    int buf = 0;
-    float ynext = 0.f, vnext = 0.f;
-    if (nnz > 0) {
-      const int col0 = indices[start];
-      if (lane < F) ynext = Y[(long long)col0 * F + lane];
-      vnext = values[start];
-    }
-    for (int c = 0; c < nnz; ++c) {
-      const float ycur = ynext;
-      const float vcur = vnext;
-      if (lane < F) ys[wave][buf][lane] = ycur;
-      wave_sync();
-      if (c + 1 < nnz) {  // issue next load during current compute
-        const int ncol = indices[start + c + 1];
-        if (lane < F) ynext = Y[(long long)ncol * F + lane];
-        vnext = values[start + c + 1];
-      }
-      float w_a, w_b;
-      if (implicit_mode) {
-        w_a = alpha * vcur;
-        w_b = 1.f + alpha * vcur;
-      } else {
-        w_a = 1.f;
-        w_b = vcur;
-      }
-      b_reg = fmaf(w_b, ycur, b_reg);
-      const float wyj = w_a * ycur;
+    // PD-deep load pipeline: a first cut double-buffered one item at a
+    // time and was latency-bound (~940 cycles/item measured at nnz=40);
+    // keeping PD row loads in flight divides the exposed HBM latency.
+    float yn[PD], vn[PD];
 #pragma unroll
-      for (int q = 0; q < F / 4; ++q) {
-        const float4 y4 = *reinterpret_cast<const float4*>(&ys[wave][buf][4 * q]);
-        acc[4 * q + 0] = fmaf(wyj, y4.x, acc[4 * q + 0]);
-        acc[4 * q + 1] = fmaf(wyj, y4.y, acc[4 * q + 1]);
-        acc[4 * q + 2] = fmaf(wyj, y4.z, acc[4 * q + 2]);
-        acc[4 * q + 3] = fmaf(wyj, y4.w, acc[4 * q + 3]);
+    for (int p = 0; p < PD; ++p) {
+      if (p < nnz) {
+        const int col = indices[start + p];
+        yn[p] = lane < F ? Y[(long long)col * F + lane] : 0.f;
+        vn[p] = values[start + p];
       }
-      buf ^= 1;
-      wave_sync();
+    }
+    for (int c0 = 0; c0 < nnz; c0 += PD) {
+#pragma unroll
+      for (int p = 0; p < PD; ++p) {
+        const int c = c0 + p;
+        if (c >= nnz) break;
+        const float ycur = yn[p];
+        const float vcur = vn[p];
+        if (lane < F) ys[wave][p][lane] = ycur;
+        wave_sync();
+        if (c + PD < nnz) {  // refill this slot PD items ahead
+          const int ncol = indices[start + c + PD];
+          if (lane < F) yn[p] = Y[(long long)ncol * F + lane];
+          vn[p] = values[start + c + PD];
+        }
+        float w_a, w_b;
+        if (implicit_mode) {
+          w_a = alpha * vcur;
+          w_b = 1.f + alpha * vcur;
+        } else {
+          w_a = 1.f;
+          w_b = vcur;
+        }
+        b_reg = fmaf(w_b, ycur, b_reg);
+        const float wyj = w_a * ycur;
+#pragma unroll
+        for (int q = 0; q < F / 4; ++q) {
+          const float4 y4 =
+              *reinterpret_cast<const float4*>(&ys[wave][p][4 * q]);
+          acc[4 * q + 0] = fmaf(wyj, y4.x, acc[4 * q + 0]);
+          acc[4 * q + 1] = fmaf(wyj, y4.y, acc[4 * q + 1]);
+          acc[4 * q + 2] = fmaf(wyj, y4.z, acc[4 * q + 2]);
+          acc[4 * q + 3] = fmaf(wyj, y4.w, acc[4 * q + 3]);
+        }
+        wave_sync();
+      }
     }
 
     // ---- YtY base (implicit) + regularization ----
