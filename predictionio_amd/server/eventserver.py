@@ -346,10 +346,12 @@ def create_app(stats_on: bool = False,
     return app
 
 
-def run(host: str = "0.0.0.0", port: int = 7070, stats_on: bool = False
-        ) -> None:
+def run(host: str = "0.0.0.0", port: int = 7070, stats_on: bool = False,
+        ssl_keyfile: str = None, ssl_certfile: str = None) -> None:
     """`pio eventserver` entry point (EventServer.scala Run.main:551-560;
-    default port 7070 as in the reference)."""
+    default port 7070 as in the reference). SSL per the reference's
+    SSLConfiguration (common/.../SSLConfiguration.scala) via uvicorn."""
     import uvicorn
     uvicorn.run(create_app(stats_on=stats_on), host=host, port=port,
-                log_level="info")
+                log_level="info", ssl_keyfile=ssl_keyfile,
+                ssl_certfile=ssl_certfile)

@@ -258,8 +258,12 @@ def create_app(config: ServerConfig,
 
 
 def run(config: ServerConfig,
-        plugins: Optional[List[EngineServerPlugin]] = None) -> None:
-    """`pio deploy` entry point (reference default port 8000)."""
+        plugins: Optional[List[EngineServerPlugin]] = None,
+        ssl_keyfile: Optional[str] = None,
+        ssl_certfile: Optional[str] = None) -> None:
+    """`pio deploy` entry point (reference default port 8000; SSL via
+    uvicorn per SSLConfiguration.scala)."""
     import uvicorn
     uvicorn.run(create_app(config, plugins), host=config.ip,
-                port=config.port, log_level="info")
+                port=config.port, log_level="info",
+                ssl_keyfile=ssl_keyfile, ssl_certfile=ssl_certfile)
