@@ -189,10 +189,13 @@ def batchpredict(engine_dir, variant, input_, output, query_partitions):
 @click.option("--ip", default="0.0.0.0")
 @click.option("--port", default=7070)
 @click.option("--stats", is_flag=True)
-def eventserver(ip, port, stats):
+@click.option("--ssl-keyfile", default=None)
+@click.option("--ssl-certfile", default=None)
+def eventserver(ip, port, stats, ssl_keyfile, ssl_certfile):
     """Launch the Event Server (EventServer.scala Run, port 7070)."""
     from predictionio_amd.server.eventserver import run
-    run(host=ip, port=port, stats_on=stats)
+    run(host=ip, port=port, stats_on=stats, ssl_keyfile=ssl_keyfile,
+        ssl_certfile=ssl_certfile)
 
 
 @cli.command()
