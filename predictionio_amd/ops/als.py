@@ -28,18 +28,21 @@ def gramian(Y: torch.Tensor) -> torch.Tensor:
 
 def woodbury_lv(Y: torch.Tensor, YtY: torch.Tensor,
                 lam: float) -> tuple:
-    """(L, V): L = chol(YtY + lam I), V = Y L^-T — the per-half-iteration
+    """(Linv, V): L = chol(YtY + lam I), Linv = L^-1, V = Y L^-T — per-half-iteration
     precompute that enables the per-row Woodbury fast path
     (als_woodbury_kernel). One fxf Cholesky + one triangular-solve GEMM
     over all items (rocBLAS trsm); the whitened rows satisfy
     v_i . v_j = y_i^T B^-1 y_j."""
     f = Y.shape[1]
-    B = YtY + lam * torch.eye(f, dtype=Y.dtype, device=Y.device)
+    eye = torch.eye(f, dtype=Y.dtype, device=Y.device)
+    B = YtY + lam * eye
     L = torch.linalg.cholesky(B)
-    # V L^T = Y  →  V = Y L^-T  (row v_i = L^-1 y_i)
-    V = torch.linalg.solve_triangular(L.mT, Y, upper=True,
-                                      left=False).contiguous()
-    return L, V
+    # Materialize L^-1 (f x f, cheap) and whiten with a GEMM: hipBLAS trsm
+    # with a 10M-row rhs hit HIPBLAS_STATUS_ALLOC_FAILED at bench scale,
+    # while hipBLASLt GEMMs of that shape are proven.
+    Linv = torch.linalg.solve_triangular(L, eye, upper=False)
+    V = (Y @ Linv.mT).contiguous()  # row v_i = L^-1 y_i
+    return Linv, V
 
 
 def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
@@ -78,13 +81,12 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
         ip, ix, vv = (indptr.contiguous(), indices.contiguous(),
                       values.contiguous())
         if implicit and pf <= 128:
-            L, V = woodbury_lv(Yp, YtYp, lam)
+            Linv, V = woodbury_lv(Yp, YtYp, lam)
             Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
                               float(alpha), True, False, 1, None)
             # X = Z L^-1 for all rows (Woodbury rows hold z; big rows get
             # overwritten by the dense pass next)
-            X = torch.linalg.solve_triangular(L, Z, upper=False, left=False)
-            X = X.contiguous()
+            X = (Z @ Linv).contiguous()
             ext.als_solve(ip, ix, vv, Yp, YtYp, None, float(lam),
                           float(alpha), True, False, 2, X)
         else:
