@@ -132,26 +132,12 @@ def run_serve(args, device):
         .to(device)
     ban_indices = bans.to(torch.int32).flatten().to(device)
 
+    from predictionio_amd.parallel.serve import sharded_topk_score
+
     def one_batch():
-        v, idx = topk_ops.topk_score(Xq, Y_local, K,
-                                     ban_indptr=ban_indptr,
-                                     ban_indices=ban_indices)
-        # merge across item shards: gather all ranks' candidates
-        if pdist.is_distributed():
-            import torch.distributed as dist
-            cand_v = torch.empty((world, B, K), dtype=v.dtype, device=device)
-            cand_i = torch.empty((world, B, K), dtype=torch.int64,
-                                 device=device)
-            dist.all_gather_into_tensor(cand_v.view(-1),
-                                        v.contiguous().view(-1))
-            gidx = torch.where(idx >= 0, idx + i_lo, idx)
-            dist.all_gather_into_tensor(cand_i.view(-1),
-                                        gidx.contiguous().view(-1))
-            allv = cand_v.permute(1, 0, 2).reshape(B, world * K)
-            alli = cand_i.permute(1, 0, 2).reshape(B, world * K)
-            mv, pos = torch.topk(allv, K, dim=1)
-            return mv, torch.gather(alli, 1, pos)
-        return v, idx
+        return sharded_topk_score(Xq, Y_local, K, item_base=i_lo,
+                                  ban_indptr=ban_indptr,
+                                  ban_indices=ban_indices)
 
     def sync():
         pdist.barrier()

@@ -176,3 +176,28 @@ class TestBenchDistPath:
     def test_bench_shard_setup(self):
         res = _spawn("_bench_shard_path", port=29619)
         assert res[0] == [16, 12] and res[1] == [16, 12]
+
+
+def _sharded_topk(rank, world):
+    """sharded_topk_score on gloo must equal single-process reference."""
+    from predictionio_amd.ops.topk import topk_score_ref
+    from predictionio_amd.parallel import dist as pdist
+    from predictionio_amd.parallel.serve import sharded_topk_score
+    g = torch.Generator().manual_seed(31)
+    B, N, f, K = 9, 40, 8, 5
+    Xq = torch.randn((B, f), generator=g).float()
+    Y = torch.randn((N, f), generator=g).float()
+    lo, hi = pdist.block_bounds(N, world, rank)
+    v, idx = sharded_topk_score(Xq, Y[lo:hi], K, item_base=lo)
+    rv, ri = topk_score_ref(Xq, Y, K)
+    assert torch.allclose(v, rv, atol=1e-5), (v - rv).abs().max()
+    # scores at chosen indices must match (ties may reorder indices)
+    chosen = (Xq @ Y.t()).gather(1, idx.clamp_min(0))
+    assert torch.allclose(chosen, rv, atol=1e-5)
+    return True
+
+
+class TestShardedServe:
+    def test_matches_reference(self):
+        res = _spawn("_sharded_topk", port=29621)
+        assert res[0] is True and res[1] is True
