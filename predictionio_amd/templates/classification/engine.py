@@ -137,6 +137,38 @@ class NaiveBayesAlgorithm(Algorithm):
                 for (i, _), l in zip(qs, labels)]
 
 
+class RandomForestAlgorithm(Algorithm):
+    """Random-forest variant (the reference's add-algorithm template adds
+    MLlib RandomForest next to NaiveBayes; here scikit-learn). Params:
+    numTrees, maxDepth, [seed]."""
+
+    def train(self, pd: PreparedData):
+        from sklearn.ensemble import RandomForestClassifier
+        X = [p.features for p in pd.labeled_points]
+        y = [p.label for p in pd.labeled_points]
+        clf = RandomForestClassifier(
+            n_estimators=int(self.params.get("numTrees", 10)),
+            max_depth=self.params.get("maxDepth"),
+            random_state=self.params.get("seed"))
+        clf.fit(X, y)
+        return clf
+
+    def predict(self, model, query) -> PredictedResult:
+        q = query if isinstance(query, Query) else Query.from_json(query)
+        label = model.predict([[q.attr0, q.attr1, q.attr2]])[0]
+        return PredictedResult(label=float(label))
+
+    def batch_predict(self, model, queries):
+        qs = [(i, q if isinstance(q, Query) else Query.from_json(q))
+              for i, q in queries]
+        if not qs:
+            return []
+        labels = model.predict(
+            [[q.attr0, q.attr1, q.attr2] for _, q in qs])
+        return [(i, PredictedResult(label=float(l)))
+                for (i, _), l in zip(qs, labels)]
+
+
 class Serving(BaseServing):
     def serve(self, query, predictions) -> PredictedResult:
         return predictions[0]
@@ -149,6 +181,7 @@ class ClassificationEngine(EngineFactory):
             data_source_class=DataSource,
             preparator_class=Preparator,
             algorithm_class={"naive": NaiveBayesAlgorithm,
+                             "randomforest": RandomForestAlgorithm,
                              "": NaiveBayesAlgorithm},
             serving_class=Serving)
 

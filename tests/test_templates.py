@@ -341,3 +341,21 @@ class TestClassificationTemplate:
                               [ce.Precision(label=0.0)]).evaluate_base(
             e, e.batch_eval([ep]))
         assert res.best_score > 0.9  # well-separated clusters
+
+
+class TestRandomForestVariant:
+    def test_rf_algorithm(self, mem_storage):
+        app_id = _mk_app(mem_storage, "MyApp3")
+        TestClassificationTemplate._seed(TestClassificationTemplate(),
+                                         mem_storage, app_id)
+        from predictionio_amd.templates.classification import engine as ce
+        from predictionio_amd.controller import EngineParams, Params
+        e = ce.ClassificationEngine.apply()
+        ep = EngineParams(
+            data_source_params=Params({"appName": "MyApp3"}),
+            algorithms_params=[("randomforest",
+                                Params({"numTrees": 10, "seed": 0}))])
+        models = e.train(ep)
+        algo = ce.RandomForestAlgorithm(ep.algorithms_params[0][1])
+        assert algo.predict(models[0], ce.Query(9, 5, 1)).label == 0.0
+        assert algo.predict(models[0], ce.Query(1, 5, 9)).label == 1.0
