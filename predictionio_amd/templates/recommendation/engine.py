@@ -92,16 +92,21 @@ class DataSource(BaseDataSource):
     [evalParams {kFold, queryNum}]."""
 
     def _read(self) -> List[Rating]:
+        """Reads `eventNames` (default rate+buy; the train-with-view
+        variant passes ["view"]); events without a rating property score
+        `implicitRating` (the reference maps buy → 4.0)."""
+        names = self.params.get("eventNames", ["rate", "buy"])
+        implicit_r = float(self.params.get("implicitRating", 4.0))
         events = event_store.find(
             app_name=self.params["appName"],
-            entity_type="user", event_names=["rate", "buy"],
+            entity_type="user", event_names=list(names),
             target_entity_type="item")
         ratings = []
         for e in events:
-            if e.event == "rate":
+            if e.event == "rate" and "rating" in e.properties:
                 r = float(e.properties.get("rating"))
             else:  # view/buy events → implicit preference weight
-                r = 4.0  # map buy event to rating value of 4 (reference)
+                r = implicit_r
             ratings.append(Rating(e.entity_id, e.target_entity_id, r))
         return ratings
 
