@@ -92,8 +92,8 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
   return X;
 }
 
-// Fused masked top-K scoring: per-(slice, wave) candidate groups; caller
-// merges with a small torch.topk. Returns (vals [B, n_slices*4*K], idx i32).
+// Fused masked top-K scoring: per-slice candidate groups; caller merges
+// with a small torch.topk. Returns (vals [B, n_slices*K], idx i32).
 std::tuple<torch::Tensor, torch::Tensor> topk_score(
     torch::Tensor Xq, torch::Tensor Y, int64_t K, int64_t n_slices,
     c10::optional<torch::Tensor> item_mask,
@@ -128,9 +128,8 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
     bi_ptr = reinterpret_cast<const long long*>(ban_indptr->data_ptr<int64_t>());
     bx_ptr = ban_indices->data_ptr<int>();
   }
-  // the kernel emits TK_WAVES=4 independent candidate groups per slice
-  auto out_val = torch::empty({B, n_slices * 4 * K}, Xq.options());
-  auto out_idx = torch::empty({B, n_slices * 4 * K},
+  auto out_val = torch::empty({B, n_slices * K}, Xq.options());
+  auto out_idx = torch::empty({B, n_slices * K},
                               Xq.options().dtype(torch::kInt32));
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();

@@ -21,11 +21,11 @@
 //    (1.3 TB/batch, 656 ms); v2's 16x-unrolled user loop blew L1I
 //    (1129 ms); v3 (lane-per-user broadcast dot) measured 293 ms — the
 //    analytic VALU bound is ~32 ms.
-//  PHASE B (reduce): scores land in an LDS tile; lane = user scans its
-//    own row slice (each wave takes a 16-item quarter) and inserts into
-//    its per-(wave,user) top-K list — no cross-lane serialization. The 4
-//    waves are independent candidate groups: output carries n_slices*4
-//    groups of K, merged host-side by one small torch.topk.
+//  PHASE B (reduce): scores land in an LDS tile; wave w owns users
+//    [16w, 16w+16) — one lane per user scans the full 64-item row and
+//    inserts into the user's single top-K list. No cross-lane or
+//    cross-wave sharing, so lists are per-user (TK_UPB*K LDS) and the
+//    output carries n_slices groups of K.
 //
 // Masks:
 //  - item_mask: optional uint8[N], 1 = globally banned (checked once per
@@ -69,13 +69,13 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
   constexpr int FP = F + 4;   // factor row stride (float4-aligned, padded)
   constexpr int SP = TK_CHUNK + 4;  // score row stride
   // dynamic LDS: xs[TK_UPB][FP] | ys[TK_CHUNK][FP] | sc[TK_UPB][SP] |
-  //              topv[TK_WAVES*TK_UPB][K] | topi[...]
+  //              topv[TK_UPB][K] | topi[...]
   extern __shared__ float lds[];
   float* xs = lds;
   float* ys = xs + TK_UPB * FP;
   float* sc = ys + TK_CHUNK * FP;
   float* topv = sc + TK_UPB * SP;
-  int* topi = reinterpret_cast<int*>(topv + TK_WAVES * TK_UPB * K);
+  int* topi = reinterpret_cast<int*>(topv + TK_UPB * K);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -102,20 +102,23 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     const int k = e % F;
     xs[u * FP + k] = (u0 + u < B) ? Xq[(u0 + u) * F + k] : 0.f;
   }
-  for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
+  for (int e = tid; e < TK_UPB * K; e += 256) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
   __syncthreads();
 
-  float th = -FLT_MAX;               // lane-as-user threshold (phase B)
-  float* tvu = topv + (wave * TK_UPB + lane) * K;
-  int* tiu = topi + (wave * TK_UPB + lane) * K;
+  // phase-B ownership: wave w, lane < 16 → user 16w + lane
+  const int u_own = wave * 16 + lane;
+  const bool owner = lane < 16;
+  float th = -FLT_MAX;               // owned user's running K-th best
+  float* tvu = topv + u_own * K;
+  int* tiu = topi + u_own * K;
 
   const int* ban = nullptr;
   int bn = 0;
-  const long long guser = u0 + lane;
-  const bool has_user = guser < B;
+  const long long guser = u0 + u_own;
+  const bool has_user = owner && guser < B;
   if (ban_indptr != nullptr && has_user) {
     const long long b0 = ban_indptr[guser];
     bn = (int)(ban_indptr[guser + 1] - b0);
@@ -147,7 +150,9 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
       const float* yrow1 = ys + (icol + 1) * FP;
       const float* yrow2 = ys + (icol + 2) * FP;
       const float* yrow3 = ys + (icol + 3) * FP;
-#pragma unroll
+      // unroll capped: full unroll kept 16 k-quads of x/y live and pushed
+      // the kernel to 334 VGPRs (1 wave/SIMD)
+#pragma unroll 2
       for (int k = 0; k < F; k += 4) {
         const f32x4 x0 = *reinterpret_cast<const f32x4*>(xrow0 + k);
         const f32x4 x1 = *reinterpret_cast<const f32x4*>(xrow1 + k);
@@ -176,11 +181,11 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
     }
     __syncthreads();
 
-    // ---- PHASE B: lane = user; this wave scans items [16w, 16w+16) ----
+    // ---- PHASE B: one lane per owned user scans the full chunk ----
     if (has_user) {
-      const int c_lo = wave * (TK_CHUNK / TK_WAVES);
-      const int c_hi = min(cn, c_lo + TK_CHUNK / TK_WAVES);
-      const float* srow = sc + lane * SP;
+      const int c_lo = 0;
+      const int c_hi = cn;
+      const float* srow = sc + u_own * SP;
       for (int c = c_lo; c < c_hi; ++c) {
         const float s = srow[c];
         if (s <= th) continue;
@@ -202,17 +207,15 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
   }
   __syncthreads();
 
-  // ---- write out: group g = slice*TK_WAVES + wave, K entries per user
-  for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
-    const int w = e / (TK_UPB * K);
-    const int u = (e / K) % TK_UPB;
+  // ---- write out: K entries per user for this slice
+  for (int e = tid; e < TK_UPB * K; e += 256) {
+    const int u = e / K;
     const int q = e % K;
     const long long gu = u0 + u;
     if (gu < B) {
-      const long long g = (long long)slice * TK_WAVES + w;
-      const long long o = (gu * n_slices * TK_WAVES + g) * K + q;
-      out_val[o] = topv[(w * TK_UPB + u) * K + q];
-      out_idx[o] = topi[(w * TK_UPB + u) * K + q];
+      const long long o = (gu * n_slices + (long long)slice) * K + q;
+      out_val[o] = topv[u * K + q];
+      out_idx[o] = topi[u * K + q];
     }
   }
 }
@@ -231,7 +234,7 @@ extern "C" void launch_topk_score(
     const size_t lds_bytes =                                                \
         sizeof(float) * ((TK_UPB + TK_CHUNK) * (FF + 4) +                   \
                          TK_UPB * (TK_CHUNK + 4)) +                         \
-        (sizeof(float) + sizeof(int)) * TK_WAVES * TK_UPB * K;              \
+        (sizeof(float) + sizeof(int)) * TK_UPB * K;                         \
     static bool attr_set_##FF = false;                                      \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
       hipFuncSetAttribute(                                                  \
