@@ -86,11 +86,15 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
   const long long u0 = (long long)ublock * TK_UPB;
 
   // phase-A tile coords: lane = ug*16 + ig → users [4*ug..+4) of this
-  // wave's 16-user group, items [4*ig..+4)
+  // wave's 16-user group, items {ig, ig+16, ig+32, ig+48}. Items stride
+  // 16 (not 4): with row stride 68 floats, rows 4 apart land on only two
+  // bank positions (68*4*4 B ≡ 16 mod 32 banks → 8-way conflict on the
+  // y b128 reads, measured 361 ms vs v3's 293); rows 16 apart spread
+  // over 8 positions (2-way).
   const int ug = lane >> 4;
   const int ig = lane & 15;
   const int urow = wave * 16 + ug * 4;   // first of this lane's 4 users
-  const int icol = ig * 4;               // first of this lane's 4 items
+  const int icol = ig;                   // lane's items: icol + 16*i
 
   const long long per = (N + n_slices - 1) / n_slices;
   const long long it0 = (long long)slice * per;
@@ -147,9 +151,9 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
       const float* xrow2 = xs + (urow + 2) * FP;
       const float* xrow3 = xs + (urow + 3) * FP;
       const float* yrow0 = ys + (icol + 0) * FP;
-      const float* yrow1 = ys + (icol + 1) * FP;
-      const float* yrow2 = ys + (icol + 2) * FP;
-      const float* yrow3 = ys + (icol + 3) * FP;
+      const float* yrow1 = ys + (icol + 16) * FP;
+      const float* yrow2 = ys + (icol + 32) * FP;
+      const float* yrow3 = ys + (icol + 48) * FP;
       // unroll capped: full unroll kept 16 k-quads of x/y live and pushed
       // the kernel to 334 VGPRs (1 wave/SIMD)
 #pragma unroll 2
@@ -173,10 +177,10 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
       // write the 4x4 tile
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        sc[(urow + 0) * SP + icol + i] = acc[i].x;
-        sc[(urow + 1) * SP + icol + i] = acc[i].y;
-        sc[(urow + 2) * SP + icol + i] = acc[i].z;
-        sc[(urow + 3) * SP + icol + i] = acc[i].w;
+        sc[(urow + 0) * SP + icol + 16 * i] = acc[i].x;
+        sc[(urow + 1) * SP + icol + 16 * i] = acc[i].y;
+        sc[(urow + 2) * SP + icol + 16 * i] = acc[i].z;
+        sc[(urow + 3) * SP + icol + 16 * i] = acc[i].w;
       }
     }
     __syncthreads();
