@@ -192,3 +192,35 @@ class TestEventStoreFacade:
             limit=2, latest=True, timeout=5.0)
         assert len(evs) == 2
         assert evs[0].event_time > evs[1].event_time
+
+
+class TestModelsBackendMatrix:
+    """Model-blob contract over every backend type (the reference runs its
+    contract suites over backend combos via env — tests/run_docker.sh)."""
+
+    @pytest.mark.parametrize("btype", ["sqlite", "localfs", "fsspec"])
+    def test_blob_contract(self, btype, tmp_path, monkeypatch):
+        from predictionio_amd.data import storage
+        from predictionio_amd.data.storage.base import Model
+        storage.reset()
+        monkeypatch.setenv("PIO_FS_BASEDIR", str(tmp_path))
+        monkeypatch.setenv("PIO_STORAGE_SOURCES_M_TYPE", btype)
+        monkeypatch.setenv("PIO_STORAGE_SOURCES_M_PATH",
+                           str(tmp_path / "blob"))
+        monkeypatch.setenv("PIO_STORAGE_SOURCES_META_TYPE", "memory")
+        for repo, src in (("METADATA", "META"), ("EVENTDATA", "META"),
+                          ("MODELDATA", "M")):
+            monkeypatch.setenv(f"PIO_STORAGE_REPOSITORIES_{repo}_SOURCE",
+                               src)
+            monkeypatch.setenv(f"PIO_STORAGE_REPOSITORIES_{repo}_NAME",
+                               repo.lower())
+        m = storage.get_model_data_models()
+        blob = bytes(range(256)) * 10
+        m.insert(Model("x1", blob))
+        assert m.get("x1").models == blob
+        m.insert(Model("x1", b"overwritten"))  # upsert semantics
+        assert m.get("x1").models == b"overwritten"
+        assert m.get("nope") is None
+        assert m.delete("x1") is True
+        assert m.delete("x1") is False
+        storage.reset()
