@@ -128,8 +128,9 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
     bi_ptr = reinterpret_cast<const long long*>(ban_indptr->data_ptr<int64_t>());
     bx_ptr = ban_indices->data_ptr<int>();
   }
-  auto out_val = torch::empty({B, n_slices * K}, Xq.options());
-  auto out_idx = torch::empty({B, n_slices * K},
+  // the kernel emits TK_WAVES=4 independent candidate groups per slice
+  auto out_val = torch::empty({B, n_slices * 4 * K}, Xq.options());
+  auto out_idx = torch::empty({B, n_slices * 4 * K},
                               Xq.options().dtype(torch::kInt32));
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
