@@ -157,6 +157,17 @@ def create_app(stats_on: bool = False,
             for p in plugins
         }}
 
+    @app.get("/plugins/{name}/{rest:path}")
+    async def plugin_rest(name: str, rest: str, request: Request):
+        """REST passthrough to a named plugin
+        (EventServer.scala:183-192 HandleREST)."""
+        for p in plugins:
+            if p.plugin_name == name:
+                return p.handle_rest({"path": rest,
+                                      "query": dict(request.query_params)})
+        return JSONResponse({"message": f"plugin {name} not found"},
+                            status_code=404)
+
     @app.post("/events.json")
     async def post_event(request: Request):
         auth = authenticate(request)

@@ -182,6 +182,17 @@ def create_app(config: ServerConfig,
             for p in plugins
         }}
 
+    @app.get("/plugins/{name}/{rest:path}")
+    async def plugin_rest(name: str, rest: str, request: Request):
+        """REST passthrough to a named plugin
+        (CreateServer.scala:656-678)."""
+        for p in plugins:
+            if p.plugin_name == name:
+                return p.handle_rest({"path": rest,
+                                      "query": dict(request.query_params)})
+        return JSONResponse({"message": f"plugin {name} not found"},
+                            status_code=404)
+
     @app.post("/queries.json")
     async def queries(request: Request):
         s = holder["st"]
