@@ -481,6 +481,14 @@ def export_cmd(appid, channel, output):
 
 
 @cli.command()
+def upgrade():
+    """Upgrade helper (Console.scala `upgrade` — prints guidance)."""
+    click.echo("[INFO] predictionio_amd is installed from this repository; "
+               "upgrade by pulling a newer revision and re-running "
+               "`pio build`.")
+
+
+@cli.command()
 @click.argument("main_class")
 @click.argument("args", nargs=-1)
 @click.option("--engine-dir", default=".")
