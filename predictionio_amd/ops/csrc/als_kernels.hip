@@ -507,63 +507,39 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     }
     wave_sync();
 
-    // ---- register/shfl Cholesky of M (lane = row; the pivot column is
-    //      broadcast with __shfl instead of LDS round trips — the LDS
-    //      version spent ~300 cycles of barrier+store/load latency per
-    //      step). Columns of L are persisted back into the M LDS rows as
-    //      a byproduct so the L^T solve can read L[k][lane]. Inner loops
-    //      guard on the UNIFORM `q < n` so unused tail iterations branch
-    //      away. ----
-    float mr[NW];
-#pragma unroll
-    for (int q = 0; q < NW; ++q)
-      mr[q] = (lane < n && q < n) ? M[wave][lane][q] : 0.f;
-    float t = lane < n ? tv[wave][lane] : 0.f;
-    wave_sync();
-    // (uniform `k < n` guards, not `break`: break blocks unrolling and
-    // the register array then needs runtime indexing)
-#pragma unroll
-    for (int k = 0; k < NW; ++k) {
-      if (k < n) {
-        float mkk = __shfl(mr[k], k);
-        mkk = mkk > 1e-30f ? mkk : 1e-30f;
-        const float dinv = rsqrtf(mkk);
-        const float ljk = lane > k ? mr[k] * dinv
-                                   : (lane == k ? mkk * dinv : 0.f);
-        mr[k] = ljk;
-        if (lane < NW) M[wave][k][lane] = ljk;  // persist column k
-#pragma unroll
-        for (int q = 0; q < NW; ++q) {
-          if (q > k && q < n) {
-            const float lqk = __shfl(ljk, q);
-            if (lane > k) mr[q] = fmaf(-ljk, lqk, mr[q]);
-          }
-        }
+    // ---- in-LDS Cholesky of M (n x n, lanes 0..n-1 = rows) ----
+    for (int k = 0; k < n; ++k) {
+      float mkk = M[wave][k][k];
+      mkk = mkk > 1e-30f ? mkk : 1e-30f;
+      const float dinv = rsqrtf(mkk);
+      if (lane == k) M[wave][k][k] = mkk * dinv;
+      else if (lane > k && lane < n) M[wave][lane][k] *= dinv;
+      wave_sync();
+      if (lane > k && lane < n) {
+        const float lik = M[wave][lane][k];
+        for (int j = k + 1; j <= lane; ++j)
+          M[wave][lane][j] = fmaf(-lik, M[wave][j][k], M[wave][lane][j]);
       }
+      wave_sync();
     }
-    // forward solve L z = rhs (z in t; L[lane][k] = mr[k])
-#pragma unroll
-    for (int k = 0; k < NW; ++k) {
-      if (k < n) {
-        const float lkk = __shfl(mr[k], k);
-        const float zk = __shfl(t, k) / lkk;
-        if (lane == k) t = zk;
-        else if (lane > k) t = fmaf(-mr[k], zk, t);
-      }
+    // forward + back substitution on tv (lane-parallel updates)
+    for (int k = 0; k < n; ++k) {
+      if (lane == k) tv[wave][k] /= M[wave][k][k];
+      wave_sync();
+      const float zk = tv[wave][k];
+      if (lane > k && lane < n)
+        tv[wave][lane] = fmaf(-M[wave][lane][k], zk, tv[wave][lane]);
+      wave_sync();
     }
-    // back solve L^T s = z: L[k][lane] = M[lane][k] (persisted rows)
-    wave_sync();
-#pragma unroll
-    for (int k = NW - 1; k >= 0; --k) {
-      if (k < n) {
-        const float lkk = __shfl(mr[k], k);
-        const float xk = __shfl(t, k) / lkk;
-        if (lane == k) t = xk;
-        else if (lane < k) t = fmaf(-M[wave][lane][k], xk, t);
-      }
+    for (int k = n - 1; k >= 0; --k) {
+      if (lane == k) tv[wave][k] /= M[wave][k][k];
+      wave_sync();
+      const float xk = tv[wave][k];
+      if (lane < k)
+        tv[wave][lane] = fmaf(-M[wave][k][lane], xk, tv[wave][lane]);
+      wave_sync();
     }
-    if (implicit_mode) t *= dv[wave][lane < n ? lane : 0];  // s = d t
-    if (lane < n) tv[wave][lane] = t;
+    if (implicit_mode && lane < n) tv[wave][lane] *= dv[wave][lane];  // s = d t
     wave_sync();
 
     // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
