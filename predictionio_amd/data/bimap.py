@@ -67,7 +67,15 @@ class BiMap:
         """Contiguous [0, n) index over distinct keys (BiMap.stringInt,
         BiMap.scala:96-120). First-seen order is preserved (the reference
         uses RDD .distinct().collect() whose order is unspecified; tests
-        must not depend on a particular assignment, only on bijectivity)."""
+        must not depend on a particular assignment, only on bijectivity).
+
+        Large inputs (the 100M-user scale of SURVEY §7 hard-part 4) go
+        through pandas.factorize — vectorized C hashing, same first-seen
+        assignment — instead of a Python dict loop."""
+        if isinstance(keys, (list, tuple, np.ndarray)) and len(keys) > 100_000:
+            import pandas as pd
+            codes, uniques = pd.factorize(np.asarray(keys))
+            return BiMap(dict(zip(uniques.tolist(), range(len(uniques)))))
         fwd: Dict[Hashable, int] = {}
         for k in keys:
             if k not in fwd:
