@@ -17,6 +17,10 @@ from predictionio_amd.data import storage
 
 def create_app() -> FastAPI:
     app = FastAPI(title="PredictionIO-AMD Dashboard")
+    # CORS helper (reference: tools/.../dashboard/CorsSupport.scala)
+    from fastapi.middleware.cors import CORSMiddleware
+    app.add_middleware(CORSMiddleware, allow_origins=["*"],
+                       allow_methods=["GET"], allow_headers=["*"])
 
     @app.get("/", response_class=HTMLResponse)
     def index():
