@@ -478,82 +478,34 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     wave_sync();
 
     // ---- M = I + D G D  (implicit)  or  G + reg I  (explicit) ----
-    // G is symmetric (G_ij = y_i^T B^-1 y_j = v_i . v_j). Computed in
-    // 4x4 PAIR TILES over the upper tile-triangle: each lane reads 4 i-
-    // rows + 4 j-rows per k-quad and produces 16 dots — 4x fewer LDS
-    // reads than the per-pair version (n^2 F/16 vs n^2 F/4 per row; the
-    // pair-wise G build measured as the stream-phase limiter: 13.4G LDS
-    // instructions ~= 21 ms per 2M-row launch at nnz 20).
+    // G is symmetric (G_ij = y_i^T B^-1 y_j): compute the upper triangle
+    // only, mirror on write. Pair index advances incrementally — no
+    // per-iteration integer division.
     const float reg = wr_scale ? lambda * (float)n : lambda;
-    const int nt = (n + 3) >> 2;              // 4-row tiles per side
-    // zero-pad the staged rows up to the tile boundary so tile reads of
-    // rows >= n contribute exactly 0
-    for (int c = n; c < nt * 4; ++c)
-      for (int e = lane; e < F; e += 64) yl[c * FP + e] = 0.f;
-    if (lane >= n && lane < nt * 4) dv[wave][lane] = 0.f;
-    wave_sync();
+    const int npairs = n * (n + 1) / 2;
     {
-      const int ntp = nt * (nt + 1) / 2;      // upper-triangle tile count
-      // map lane → first tile (ti, tj), tj >= ti
-      int p = lane, ti = 0;
-      while (p >= nt - ti && ti < nt) { p -= nt - ti; ++ti; }
-      int tj = ti + p;
-      for (int pp = lane; pp < ntp; pp += 64) {
-        const int i0 = ti * 4, j0 = tj * 4;
-        f32x4_t acc[4];                        // acc[a].b = v_{i0+a}.v_{j0+b}
+      // map lane → first (i, j) with j >= i in the flattened triangle
+      int p = lane, i = 0;
+      while (p >= n - i && i < n) { p -= n - i; ++i; }
+      int j = i + p;
+      for (int pp = lane; pp < npairs; pp += 64) {
+        const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
+        const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
+        f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};  // 2x v_pk_fma_f32 per q
 #pragma unroll
-        for (int a = 0; a < 4; ++a) acc[a] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-#pragma unroll 2
-        for (int q = 0; q < F / 4; ++q) {
-          f32x4_t yiq[4], yjq[4];
-#pragma unroll
-          for (int a = 0; a < 4; ++a) {
-            yiq[a] = *reinterpret_cast<const f32x4_t*>(
-                &yl[(i0 + a) * FP + 4 * q]);
-            yjq[a] = *reinterpret_cast<const f32x4_t*>(
-                &yl[(j0 + a) * FP + 4 * q]);
-          }
-#pragma unroll
-          for (int a = 0; a < 4; ++a) {
-            // dot(v_{i0+a}, v_{j0+b}) partial: elementwise mul then the
-            // horizontal sum is deferred via per-component accumulation
-            acc[a].x += yiq[a].x * yjq[0].x + yiq[a].y * yjq[0].y
-                      + yiq[a].z * yjq[0].z + yiq[a].w * yjq[0].w;
-            acc[a].y += yiq[a].x * yjq[1].x + yiq[a].y * yjq[1].y
-                      + yiq[a].z * yjq[1].z + yiq[a].w * yjq[1].w;
-            acc[a].z += yiq[a].x * yjq[2].x + yiq[a].y * yjq[2].y
-                      + yiq[a].z * yjq[2].z + yiq[a].w * yjq[2].w;
-            acc[a].w += yiq[a].x * yjq[3].x + yiq[a].y * yjq[3].y
-                      + yiq[a].z * yjq[3].z + yiq[a].w * yjq[3].w;
-          }
+        for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
+        float dot = acc4.x + acc4.y + acc4.z + acc4.w;
+        if (implicit_mode) {
+          dot *= dv[wave][i] * dv[wave][j];
+          if (i == j) dot += 1.f;
+        } else if (i == j) {
+          dot += reg;
         }
-        // scale + write the tile (and its mirror)
-#pragma unroll
-        for (int a = 0; a < 4; ++a) {
-          const int i = i0 + a;
-          if (i < n) {
-            const float di = implicit_mode ? dv[wave][i] : 1.f;
-            float vals[4] = {acc[a].x, acc[a].y, acc[a].z, acc[a].w};
-#pragma unroll
-            for (int b = 0; b < 4; ++b) {
-              const int j = j0 + b;
-              if (j < n) {
-                float dot = vals[b];
-                if (implicit_mode) {
-                  dot *= di * dv[wave][j];
-                  if (i == j) dot += 1.f;
-                } else if (i == j) {
-                  dot += reg;
-                }
-                M[wave][i][j] = dot;
-                M[wave][j][i] = dot;
-              }
-            }
-          }
-        }
-        // advance 64 tile slots
-        tj += 64;
-        while (tj >= nt && ti < nt) { ++ti; tj -= nt - ti; }
+        M[wave][i][j] = dot;
+        M[wave][j][i] = dot;
+        // advance 64 triangle slots
+        j += 64;
+        while (j >= n && i < n) { ++i; j -= n - i; }
       }
     }
     wave_sync();
