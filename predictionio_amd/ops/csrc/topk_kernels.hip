@@ -36,7 +36,7 @@
 #include <float.h>
 #include <hip/hip_runtime.h>
 
-#define TK_CHUNK 64   // items staged per LDS pass
+#define TK_CHUNK 32   // items staged per LDS pass (32: fits 3 WGs/CU with K=20 lists)
 #define TK_MAXK 64    // max supported K
 #define TK_WAVES 4
 #define TK_UPB 64     // users per block (= lane count)
@@ -55,7 +55,7 @@ __device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
 }
 
 template <int F>
-__global__ __launch_bounds__(256) void topk_score_kernel(
+__global__ __launch_bounds__(256, 3) void topk_score_kernel(
     const float* __restrict__ Xq,        // B x F query vectors
     const float* __restrict__ Y,         // N x F item factors
     const uint8_t* __restrict__ item_mask,       // N or nullptr
@@ -87,7 +87,11 @@ __global__ __launch_bounds__(256) void topk_score_kernel(
   const long long it0 = (long long)slice * per;
   const long long it1 = min(N, it0 + per);
 
-  // ---- load query vectors: coalesced stage into ys, transpose to regs
+  // ---- load query vectors: coalesced stage into ys, transpose to regs.
+  // NOTE: TK_UPB (64) rows are staged but ys is TK_CHUNK (32) rows — the
+  // upper rows intentionally overflow into the top-list region, which is
+  // initialized only AFTER the transpose below, and the total dynamic
+  // allocation covers it (x staging 16.9 KB < 49.4 KB).
   for (int e = tid; e < TK_UPB * F; e += 256) {
     const int u = e / F;
     const int k = e % F;
