@@ -166,15 +166,12 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
       const float dinv = diag_inv;
       for (int j = k + 1 + tid; j < F; j += 256) As[j][k] *= dinv;
       __syncthreads();
-      // trailing update: A[i][j] -= L[i][k] * L[j][k] for i>=j>k.
-      // Iterate the full FxF grid with COMPILE-TIME divisor (shifts) and
-      // predicate: an earlier version divided by the runtime `rem`
-      // (~20 instr per element) and ran 4x slower at F=128.
-      for (int e = tid; e < F * F; e += 256) {
-        const int i = e / F;
-        const int j = e % F;
-        if (i > k && j > k && j <= i)
-          As[i][j] = fmaf(-As[i][k], As[j][k], As[i][j]);
+      // trailing update: A[i][j] -= L[i][k] * L[j][k] for i>=j>k
+      const int rem = F - k - 1;
+      for (int e = tid; e < rem * rem; e += 256) {
+        const int i = k + 1 + e / rem;
+        const int j = k + 1 + e % rem;
+        if (j <= i) As[i][j] = fmaf(-As[i][k], As[j][k], As[i][j]);
       }
       __syncthreads();
     }
