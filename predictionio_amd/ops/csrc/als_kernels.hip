@@ -249,8 +249,9 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
   const int wave = threadIdx.x >> 6;
 
   constexpr int PD = 4;  // global-load pipeline depth (latency hiding)
+  constexpr int LP = F + 4;  // Lc row stride: 16B-aligned for b128 reads
   __shared__ float ys[2][PD][F];       // staged y slots per wave
-  __shared__ float Lc[2][F][F + 1];    // persisted L columns: Lc[w][k][j] = L[j][k]
+  __shared__ float Lc[2][F][LP];       // persisted L columns: Lc[w][k][j] = L[j][k]
 
   for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
        row += (long long)gridDim.x * 2) {
@@ -344,13 +345,24 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
       const float ljk = lane > k ? acc[k] * dinv
                                  : (lane == k ? lkk * dinv : 0.f);
       // persist column k: Lc[k][j] = L[j][k]. Guard: for F < 64 lanes
-      // >= F would write past the (F+1)-wide row into the next column.
+      // >= F would write past the row into the next column.
       if (lane < F) Lc[wave][k][lane] = ljk;
       wave_sync();
+      // trailing update reading the pivot column in b128 QUADS: the
+      // scalar version compiled to ds_read_b32 + s_waitcnt lgkmcnt(0)
+      // per element — a full LDS round trip exposed F-k times per step
+      // (seen in the ISA; the fixed per-row solve cost dominated the
+      // kernel). One quad read amortizes the wait over 4 fmas.
 #pragma unroll
-      for (int m = k + 1; m < F; ++m) {
-        const float lmk = Lc[wave][k][m];  // broadcast read
-        if (lane > k) acc[m] = fmaf(-ljk, lmk, acc[m]);
+      for (int m4 = (k + 1) & ~3; m4 < F; m4 += 4) {
+        const float4 lq =
+            *reinterpret_cast<const float4*>(&Lc[wave][k][m4]);
+        if (lane > k) {
+          if (m4 + 0 > k) acc[m4 + 0] = fmaf(-ljk, lq.x, acc[m4 + 0]);
+          if (m4 + 1 > k) acc[m4 + 1] = fmaf(-ljk, lq.y, acc[m4 + 1]);
+          if (m4 + 2 > k) acc[m4 + 2] = fmaf(-ljk, lq.z, acc[m4 + 2]);
+          if (m4 + 3 > k) acc[m4 + 3] = fmaf(-ljk, lq.w, acc[m4 + 3]);
+        }
       }
       wave_sync();
     }

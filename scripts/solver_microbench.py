@@ -50,3 +50,6 @@ if __name__ == "__main__":
     for f in (64, 128):
         for nnz_per_row in (5, 20, 40):
             bench(n_rows, n_cols, f, nnz_per_row)
+    # the 8-GPU item-side shape: rows densify with world size
+    for nnz_per_row in (100, 200):
+        bench(n_rows // 4, n_cols, 64, nnz_per_row)
