@@ -255,3 +255,128 @@ class TestPlugins:
         r = c.post("/events.json?accessKey=K", json=dict(EV, event="view"))
         assert r.status_code == 403
         storage.reset()
+
+
+class TestSegmentIOTypes:
+    """All six Segment.io message types (SegmentIOConnector.scala:52-92)."""
+
+    def _post(self, client, payload):
+        return client.post("/webhooks/segmentio.json?accessKey=SECRET",
+                           json=payload)
+
+    def test_identify(self, client):
+        r = self._post(client, {"version": "2", "type": "identify",
+                                "user_id": "u1",
+                                "traits": {"email": "a@b.c"}})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=identify"
+                       ).json()[0]
+        assert e["properties"]["traits"]["email"] == "a@b.c"
+
+    def test_page_screen(self, client):
+        for typ in ("page", "screen"):
+            r = self._post(client, {"version": "2", "type": typ,
+                                    "user_id": "u1", "name": "home"})
+            assert r.status_code == 201
+            e = client.get(f"/events.json?accessKey=SECRET&event={typ}"
+                           ).json()[0]
+            assert e["properties"]["name"] == "home"
+
+    def test_alias(self, client):
+        r = self._post(client, {"version": "2", "type": "alias",
+                                "user_id": "u2", "previous_id": "anon9"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=alias").json()[0]
+        assert e["properties"]["previous_id"] == "anon9"
+
+    def test_group(self, client):
+        r = self._post(client, {"version": "2", "type": "group",
+                                "user_id": "u3", "group_id": "g7",
+                                "traits": {"plan": "pro"}})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=group").json()[0]
+        assert e["properties"]["group_id"] == "g7"
+
+    def test_anonymous_id_fallback(self, client):
+        r = self._post(client, {"version": "2", "type": "track",
+                                "anonymous_id": "anon1", "event": "x"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=track").json()[0]
+        assert e["entityId"] == "anon1"
+
+    def test_context_folded_into_properties(self, client):
+        r = self._post(client, {"version": "2", "type": "track",
+                                "user_id": "u4", "event": "buy",
+                                "context": {"ip": "1.2.3.4"},
+                                "properties": {"sku": 9}})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=track").json()[0]
+        assert e["properties"]["context"]["ip"] == "1.2.3.4"
+
+    def test_missing_version(self, client):
+        assert self._post(client, {"type": "track", "user_id": "u",
+                                   "event": "x"}).status_code == 400
+
+    def test_unknown_type(self, client):
+        assert self._post(client, {"version": "2", "type": "nope",
+                                   "user_id": "u"}).status_code == 400
+
+
+class TestMailChimpTypes:
+    def _post(self, client, form):
+        return client.post("/webhooks/mailchimp.form?accessKey=SECRET",
+                           data=form)
+
+    def test_unsubscribe(self, client):
+        r = self._post(client, {
+            "type": "unsubscribe", "fired_at": "2026-03-26 21:40:57",
+            "data[action]": "unsub", "data[reason]": "manual",
+            "data[id]": "8a25ff1d98", "data[list_id]": "a6b5da1054",
+            "data[email]": "x@y.z", "data[email_type]": "html",
+            "data[merges][EMAIL]": "x@y.z", "data[merges][FNAME]": "X",
+            "data[merges][LNAME]": "Y", "data[ip_opt]": "1.1.1.1",
+            "data[campaign_id]": "cb398d21d2"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=unsubscribe"
+                       ).json()[0]
+        assert e["properties"]["action"] == "unsub"
+
+    def test_upemail(self, client):
+        r = self._post(client, {
+            "type": "upemail", "fired_at": "2026-03-26 22:15:09",
+            "data[list_id]": "a6b5da1054", "data[new_id]": "51da8c3259",
+            "data[new_email]": "new@x.y", "data[old_email]": "old@x.y"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=upemail"
+                       ).json()[0]
+        assert e["entityId"] == "51da8c3259"
+
+    def test_cleaned(self, client):
+        r = self._post(client, {
+            "type": "cleaned", "fired_at": "2026-03-26 22:01:00",
+            "data[list_id]": "a6b5da1054", "data[campaign_id]": "c1",
+            "data[reason]": "hard", "data[email]": "gone@x.y"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=cleaned"
+                       ).json()[0]
+        assert e["entityType"] == "list"
+
+    def test_campaign(self, client):
+        r = self._post(client, {
+            "type": "campaign", "fired_at": "2026-03-26 21:31:21",
+            "data[id]": "5aa2102003", "data[subject]": "S",
+            "data[status]": "sent", "data[reason]": "",
+            "data[list_id]": "a6b5da1054"})
+        assert r.status_code == 201
+        e = client.get("/events.json?accessKey=SECRET&event=campaign"
+                       ).json()[0]
+        assert e["entityType"] == "campaign"
+
+    def test_missing_type(self, client):
+        assert self._post(client, {"fired_at": "2026-01-01 00:00:00"
+                                   }).status_code == 400
+
+    def test_missing_field(self, client):
+        assert self._post(client, {"type": "subscribe",
+                                   "fired_at": "2026-01-01 00:00:00"
+                                   }).status_code == 400
