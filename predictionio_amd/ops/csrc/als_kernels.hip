@@ -420,7 +420,7 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 template <int F>
-__global__ __launch_bounds__(128) void als_woodbury_kernel(
+__global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
     const float* __restrict__ values,
@@ -439,10 +439,10 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
   const int wave = threadIdx.x >> 6;
 
   __shared__ float Yl[2][NW][FP];      // staged factor rows (Y or V)
-  __shared__ float M[2][NW][NW + 1];   // I+DGD (implicit) / G+regI (explicit)
-  __shared__ float tv[2][NW];          // rhs, then solution t
+  constexpr int MP = NW + 4;  // M row stride: 16B-aligned for b128 reads
+  __shared__ float M[2][NW][MP];       // I+DGD (implicit) / G+regI (explicit)
+  __shared__ float tv[2][NW];          // col ids, then rhs, then solution
   __shared__ float dv[2][NW];          // D diagonal (implicit)
-  __shared__ int cols_l[2][NW];        // prefetched column ids
 
   float* yl = &Yl[wave][0][0];
   const float* src = implicit_mode ? V : Y;
@@ -462,12 +462,14 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     // One coalesced load of the column ids first, THEN the factor-row
     // loads — with `indices[start+c]` inline each row load waited on its
     // own index fetch (PMC: 7166 wait vs 599 busy cycles per wave); via
-    // LDS all n row loads issue back-to-back. (Strided over e so F = 128
-    // works with 64 lanes.)
-    if (lane < n) cols_l[wave][lane] = indices[start + lane];
+    // LDS all n row loads issue back-to-back. The ids are staged through
+    // the tv buffer (bit-cast; tv's real use starts after the last id
+    // read). (Strided over e so F = 128 works with 64 lanes.)
+    if (lane < n)
+      tv[wave][lane] = __int_as_float(indices[start + lane]);
     wave_sync();
     for (int c = 0; c < n; ++c) {
-      const long long col = cols_l[wave][c];
+      const long long col = __float_as_int(tv[wave][c]);
       for (int e = lane; e < F; e += 64)
         yl[c * FP + e] = src[col * F + e];
     }
@@ -519,39 +521,74 @@ __global__ __launch_bounds__(128) void als_woodbury_kernel(
     }
     wave_sync();
 
-    // ---- in-LDS Cholesky of M (n x n, lanes 0..n-1 = rows) ----
-    for (int k = 0; k < n; ++k) {
-      float mkk = M[wave][k][k];
-      mkk = mkk > 1e-30f ? mkk : 1e-30f;
-      const float dinv = rsqrtf(mkk);
-      if (lane == k) M[wave][k][k] = mkk * dinv;
-      else if (lane > k && lane < n) M[wave][lane][k] *= dinv;
-      wave_sync();
-      if (lane > k && lane < n) {
-        const float lik = M[wave][lane][k];
-        for (int j = k + 1; j <= lane; ++j)
-          M[wave][lane][j] = fmaf(-lik, M[wave][j][k], M[wave][lane][j]);
+    // ---- Cholesky of M: register rows + quad pivot-column reads ----
+    // lane = row of M in registers (mr); each finished column is written
+    // into M's row k (M[k][j] = L[j][k]) and the trailing update reads it
+    // back in b128 QUADS — the pure-LDS version compiled to one
+    // ds_read_b32 + s_waitcnt per element (the same exposed-round-trip
+    // disease fixed in the dense wave kernel, 1.7x there). A pure-shfl
+    // variant without the quad reads measured slower (528 shfls); this
+    // hybrid keeps the accumulators in registers and amortizes the LDS
+    // latency 4-wide. Uniform `k < n` guards (no break) keep the k-loops
+    // unrollable so mr[] stays in registers.
+    float mr[NW];
+#pragma unroll
+    for (int q = 0; q < NW; ++q)
+      mr[q] = (lane < n && q < n) ? M[wave][lane][q] : 0.f;
+    float t = lane < n ? tv[wave][lane] : 0.f;
+    wave_sync();
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {
+        float mkk = __shfl(mr[k], k);
+        mkk = mkk > 1e-30f ? mkk : 1e-30f;
+        const float dinv = rsqrtf(mkk);
+        const float ljk = lane > k ? mr[k] * dinv
+                                   : (lane == k ? mkk * dinv : 0.f);
+        mr[k] = ljk;
+        if (lane < NW) M[wave][k][lane] = ljk;  // column k stored as row k
+        wave_sync();
+#pragma unroll
+        for (int j4 = (k + 1) & ~3; j4 < NW; j4 += 4) {
+          if (j4 < n) {
+            const float4 q4 =
+                *reinterpret_cast<const float4*>(&M[wave][k][j4]);
+            if (lane > k) {
+              // elements with j >= n update never-read mr slots (safe)
+              if (j4 + 0 > k) mr[j4 + 0] = fmaf(-ljk, q4.x, mr[j4 + 0]);
+              if (j4 + 1 > k) mr[j4 + 1] = fmaf(-ljk, q4.y, mr[j4 + 1]);
+              if (j4 + 2 > k) mr[j4 + 2] = fmaf(-ljk, q4.z, mr[j4 + 2]);
+              if (j4 + 3 > k) mr[j4 + 3] = fmaf(-ljk, q4.w, mr[j4 + 3]);
+            }
+          }
+        }
+        wave_sync();
       }
-      wave_sync();
     }
-    // forward + back substitution on tv (lane-parallel updates)
-    for (int k = 0; k < n; ++k) {
-      if (lane == k) tv[wave][k] /= M[wave][k][k];
-      wave_sync();
-      const float zk = tv[wave][k];
-      if (lane > k && lane < n)
-        tv[wave][lane] = fmaf(-M[wave][lane][k], zk, tv[wave][lane]);
-      wave_sync();
+    // forward solve L z = rhs (L[lane][k] = mr[k], registers)
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], k);
+        const float zk = __shfl(t, k) / lkk;
+        if (lane == k) t = zk;
+        else if (lane > k) t = fmaf(-mr[k], zk, t);
+      }
     }
-    for (int k = n - 1; k >= 0; --k) {
-      if (lane == k) tv[wave][k] /= M[wave][k][k];
-      wave_sync();
-      const float xk = tv[wave][k];
-      if (lane < k)
-        tv[wave][lane] = fmaf(-M[wave][k][lane], xk, tv[wave][lane]);
-      wave_sync();
+    // back solve L^T s = z: L[k][lane] = M[lane][k] (row `lane` holds
+    // column `lane`)
+    wave_sync();
+#pragma unroll
+    for (int k = NW - 1; k >= 0; --k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], k);
+        const float xk = __shfl(t, k) / lkk;
+        if (lane == k) t = xk;
+        else if (lane < k) t = fmaf(-M[wave][lane][k], xk, t);
+      }
     }
-    if (implicit_mode && lane < n) tv[wave][lane] *= dv[wave][lane];  // s = d t
+    if (implicit_mode) t *= dv[wave][lane < n ? lane : 0];  // s = d t
+    if (lane < n) tv[wave][lane] = t;
     wave_sync();
 
     // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
