@@ -521,63 +521,74 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     }
     wave_sync();
 
-    // ---- Cholesky of M in LDS with b128 quads both ways ----
-    // Runtime k-loop (small code): the register-row variant's fully
-    // unrolled k/j loops produced a 34 KB straight-line body — beyond the
-    // 32 KB L1I, so every row streamed its own instructions from L2
-    // (measured as the dominant fixed cost). Here the trailing update
-    // reads the scaled pivot column from row k's upper mirror (dead G
-    // values after the build) and RMWs the lane's own row, both as
-    // 16B-aligned quads (stride MP = NW+4), amortizing the LDS round
-    // trips 4-wide with ~1.5 KB of code.
-    for (int k = 0; k < n; ++k) {
-      float mkk = M[wave][k][k];
-      mkk = mkk > 1e-30f ? mkk : 1e-30f;
-      const float dinv = rsqrtf(mkk);
-      if (lane >= k && lane < n) {
-        const float sc = lane == k ? mkk * dinv : M[wave][lane][k] * dinv;
-        M[wave][lane][k] = sc;   // column k (lower)
-        M[wave][k][lane] = sc;   // mirror into row k (upper; coalesced)
-      }
-      wave_sync();
-      if (lane > k && lane < n) {
-        const float lik = M[wave][lane][k];
-        for (int j4 = (k + 1) & ~3; j4 <= lane; j4 += 4) {
-          const float4 col =
-              *reinterpret_cast<const float4*>(&M[wave][k][j4]);
-          float4 row = *reinterpret_cast<float4*>(&M[wave][lane][j4]);
-          if (j4 + 0 > k && j4 + 0 <= lane)
-            row.x = fmaf(-lik, col.x, row.x);
-          if (j4 + 1 > k && j4 + 1 <= lane)
-            row.y = fmaf(-lik, col.y, row.y);
-          if (j4 + 2 > k && j4 + 2 <= lane)
-            row.z = fmaf(-lik, col.z, row.z);
-          if (j4 + 3 > k && j4 + 3 <= lane)
-            row.w = fmaf(-lik, col.w, row.w);
-          *reinterpret_cast<float4*>(&M[wave][lane][j4]) = row;
+    // ---- Cholesky of M: register rows + quad pivot-column reads ----
+    // lane = row of M in registers (mr); each finished column is written
+    // into M's row k (M[k][j] = L[j][k]) and the trailing update reads it
+    // back in b128 QUADS — the pure-LDS version compiled to one
+    // ds_read_b32 + s_waitcnt per element (the same exposed-round-trip
+    // disease fixed in the dense wave kernel, 1.7x there). A pure-shfl
+    // variant without the quad reads measured slower (528 shfls); this
+    // hybrid keeps the accumulators in registers and amortizes the LDS
+    // latency 4-wide. Uniform `k < n` guards (no break) keep the k-loops
+    // unrollable so mr[] stays in registers.
+    float mr[NW];
+#pragma unroll
+    for (int q = 0; q < NW; ++q)
+      mr[q] = (lane < n && q < n) ? M[wave][lane][q] : 0.f;
+    float t = lane < n ? tv[wave][lane] : 0.f;
+    wave_sync();
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {
+        float mkk = __shfl(mr[k], k);
+        mkk = mkk > 1e-30f ? mkk : 1e-30f;
+        const float dinv = rsqrtf(mkk);
+        const float ljk = lane > k ? mr[k] * dinv
+                                   : (lane == k ? mkk * dinv : 0.f);
+        mr[k] = ljk;
+        if (lane < NW) M[wave][k][lane] = ljk;  // column k stored as row k
+        wave_sync();
+#pragma unroll
+        for (int j4 = (k + 1) & ~3; j4 < NW; j4 += 4) {
+          if (j4 < n) {
+            const float4 q4 =
+                *reinterpret_cast<const float4*>(&M[wave][k][j4]);
+            if (lane > k) {
+              // elements with j >= n update never-read mr slots (safe)
+              if (j4 + 0 > k) mr[j4 + 0] = fmaf(-ljk, q4.x, mr[j4 + 0]);
+              if (j4 + 1 > k) mr[j4 + 1] = fmaf(-ljk, q4.y, mr[j4 + 1]);
+              if (j4 + 2 > k) mr[j4 + 2] = fmaf(-ljk, q4.z, mr[j4 + 2]);
+              if (j4 + 3 > k) mr[j4 + 3] = fmaf(-ljk, q4.w, mr[j4 + 3]);
+            }
+          }
         }
+        wave_sync();
       }
-      wave_sync();
     }
-    // forward + back substitution (lane-parallel; L[lane][k] = M[lane][k],
-    // L[k][lane] = M[lane][k] for k > lane via the row mirror)
-    for (int k = 0; k < n; ++k) {
-      if (lane == k) tv[wave][k] /= M[wave][k][k];
-      wave_sync();
-      const float zk = tv[wave][k];
-      if (lane > k && lane < n)
-        tv[wave][lane] = fmaf(-M[wave][lane][k], zk, tv[wave][lane]);
-      wave_sync();
+    // forward solve L z = rhs (L[lane][k] = mr[k], registers)
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], k);
+        const float zk = __shfl(t, k) / lkk;
+        if (lane == k) t = zk;
+        else if (lane > k) t = fmaf(-mr[k], zk, t);
+      }
     }
-    for (int k = n - 1; k >= 0; --k) {
-      if (lane == k) tv[wave][k] /= M[wave][k][k];
-      wave_sync();
-      const float xk = tv[wave][k];
-      if (lane < k)
-        tv[wave][lane] = fmaf(-M[wave][lane][k], xk, tv[wave][lane]);
-      wave_sync();
+    // back solve L^T s = z: L[k][lane] = M[lane][k] (row `lane` holds
+    // column `lane`)
+    wave_sync();
+#pragma unroll
+    for (int k = NW - 1; k >= 0; --k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], k);
+        const float xk = __shfl(t, k) / lkk;
+        if (lane == k) t = xk;
+        else if (lane < k) t = fmaf(-M[wave][lane][k], xk, t);
+      }
     }
-    if (implicit_mode && lane < n) tv[wave][lane] *= dv[wave][lane];
+    if (implicit_mode) t *= dv[wave][lane < n ? lane : 0];  // s = d t
+    if (lane < n) tv[wave][lane] = t;
     wave_sync();
 
     // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
