@@ -167,3 +167,50 @@ class TestTopKRef:
         ref = sum(F.cosine_similarity(Y[q].unsqueeze(0), Y) for q in qitems)
         rv, ri = torch.topk(ref, 5)
         assert idxs[0].tolist() == ri.tolist()
+
+
+class TestCosineTopK:
+    def test_matches_manual_cosine(self):
+        import torch
+        from predictionio_amd.ops.topk import cosine_topk, normalize_rows
+        g = torch.Generator().manual_seed(3)
+        Y = torch.randn((50, 8), generator=g)
+        Yn = normalize_rows(Y)
+        q = Yn[3] + Yn[7]
+        v, idx = cosine_topk(q, Yn, 5)
+        manual = (q.unsqueeze(0) @ Yn.t()).squeeze(0)
+        mv, mi = manual.topk(5)
+        assert torch.allclose(v[0], mv, atol=1e-5)
+        assert set(idx[0].tolist()) == set(mi.tolist())
+
+    def test_normalize_rows_zero_safe(self):
+        import torch
+        from predictionio_amd.ops.topk import normalize_rows
+        Y = torch.zeros((3, 4))
+        assert torch.isfinite(normalize_rows(Y)).all()
+
+
+class TestPypio:
+    def test_save_load_model(self, mem_storage):
+        from predictionio_amd import pypio
+        pypio.init()
+        iid = pypio.save_model({"weights": [1, 2, 3]})
+        assert pypio.load_model(iid) == {"weights": [1, 2, 3]}
+        inst = mem_storage.get_meta_data_engine_instances().get(iid)
+        assert inst.status == "COMPLETED"
+
+    def test_find_events_columns(self, mem_storage):
+        from predictionio_amd import pypio
+        from predictionio_amd.data.events import DataMap, Event, utcnow
+        from predictionio_amd.data.storage.base import App
+        app_id = mem_storage.get_meta_data_apps().insert(App(0, "pyapp"))
+        mem_storage.get_l_events().init(app_id)
+        mem_storage.get_l_events().insert(
+            Event(event="rate", entity_type="user", entity_id="u1",
+                  target_entity_type="item", target_entity_id="i1",
+                  properties=DataMap({"rating": 5}),
+                  event_time=utcnow()), app_id)
+        evs = pypio.find_events("pyapp")
+        cols = pypio.events_to_columns(evs)
+        assert cols["entityId"] == ["u1"]
+        assert cols["properties"][0]["rating"] == 5
