@@ -126,3 +126,19 @@ class TestFeedback:
                 break
             time.sleep(0.05)
         assert posted and "/events.json" in posted[0].full_url
+
+
+class TestStopAuth:
+    def test_stop_requires_key(self, mem_storage, monkeypatch):
+        _train()
+        from predictionio_amd.server.queryserver import (
+            ServerConfig, create_app,
+        )
+        c = TestClient(create_app(ServerConfig(
+            engine_factory=FACTORY, access_key="SECRET")))
+        r = c.post("/stop")
+        assert r.status_code == 401
+        # correct key accepted (patch os.kill so the test survives)
+        import os
+        monkeypatch.setattr(os, "kill", lambda *a: None)
+        assert c.post("/stop?accessKey=SECRET").status_code == 200
