@@ -201,3 +201,58 @@ class TestShardedServe:
     def test_matches_reference(self):
         res = _spawn("_sharded_topk", port=29621)
         assert res[0] is True and res[1] is True
+
+
+def _dist_als_implicit(rank, world):
+    """Implicit-mode distributed ALS on 2 ranks equals single-process."""
+    from predictionio_amd.models.als import ALSParams, ALSTrainer
+    from predictionio_amd.ops import als as als_ops
+    g = torch.Generator().manual_seed(77)
+    n_users, n_items, f = 30, 20, 16
+    nnz = 300
+    users = torch.randint(0, n_users, (nnz,), generator=g, dtype=torch.int32)
+    items = torch.randint(0, n_items, (nnz,), generator=g, dtype=torch.int32)
+    vals = torch.ones(nnz)
+    users, items, vals = als_ops.aggregate_ratings(users, items, vals,
+                                                   n_items, "sum")
+    p = ALSParams(rank=f, iterations=2, lambda_=0.05, alpha=10.0,
+                  implicit=True, seed=0)
+    t = ALSTrainer(p, n_users, n_items, torch.device("cpu"))
+    t.set_ratings(users, items, vals)
+    gen = torch.Generator().manual_seed(5)
+    X0 = torch.randn((n_users, f), generator=gen) / (f ** 0.5)
+    Y0 = torch.randn((n_items, f), generator=gen) / (f ** 0.5)
+    t.X = X0[t.u_lo:t.u_hi].clone()
+    t.Y = Y0[t.i_lo:t.i_hi].clone()
+    for _ in range(p.iterations):
+        t.step()
+    X, Y = t.gather_factors()
+    return (X.numpy().tolist(), Y.numpy().tolist())
+
+
+class TestDistributedImplicit:
+    def test_matches_single(self):
+        res = _spawn("_dist_als_implicit", port=29623)
+        assert res[0] == res[1]
+        from predictionio_amd.models.als import ALSParams, ALSTrainer
+        from predictionio_amd.ops import als as als_ops
+        g = torch.Generator().manual_seed(77)
+        n_users, n_items, f = 30, 20, 16
+        users = torch.randint(0, n_users, (300,), generator=g,
+                              dtype=torch.int32)
+        items = torch.randint(0, n_items, (300,), generator=g,
+                              dtype=torch.int32)
+        vals = torch.ones(300)
+        users, items, vals = als_ops.aggregate_ratings(users, items, vals,
+                                                       n_items, "sum")
+        p = ALSParams(rank=f, iterations=2, lambda_=0.05, alpha=10.0,
+                      implicit=True, seed=0)
+        t = ALSTrainer(p, n_users, n_items, torch.device("cpu"))
+        t.set_ratings(users, items, vals)
+        gen = torch.Generator().manual_seed(5)
+        t.X = torch.randn((n_users, f), generator=gen) / (f ** 0.5)
+        t.Y = torch.randn((n_items, f), generator=gen) / (f ** 0.5)
+        for _ in range(p.iterations):
+            t.step()
+        assert torch.allclose(torch.tensor(res[0][0]), t.X,
+                              atol=1e-4, rtol=1e-4)
