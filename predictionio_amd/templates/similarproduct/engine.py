@@ -103,7 +103,11 @@ class DataSource(BaseDataSource):
     """Params: appName."""
 
     def read_training(self) -> TrainingData:
+        """Params: appName, [eventNames] — the train-with-rate-event
+        variant passes ["rate"] (or mixes) and the events still feed the
+        implicit trainer as unit preferences."""
         app = self.params["appName"]
+        names = list(self.params.get("eventNames", ["view"]))
         users = {eid: pm.to_dict() for eid, pm in
                  event_store.aggregate_properties(app, "user").items()}
         items = {eid: Item(categories=pm.get_opt("categories"))
@@ -113,7 +117,7 @@ class DataSource(BaseDataSource):
             ViewEvent(e.entity_id, e.target_entity_id,
                       e.event_time.timestamp())
             for e in event_store.find(app, entity_type="user",
-                                      event_names=["view"],
+                                      event_names=names,
                                       target_entity_type="item")
         ]
         return TrainingData(users, items, views)

@@ -105,3 +105,10 @@ class TestTrainDeployFlow:
                                 "tests.fake_engine_eval.ZeroEvaluation"])
         assert r.exit_code == 0, r.output
         assert "Best score" in r.output
+
+
+class TestStatus:
+    def test_status_healthy(self, runner):
+        r = runner.invoke(cli, ["status"])
+        assert r.exit_code == 0, r.output
+        assert "ready to go" in r.output
