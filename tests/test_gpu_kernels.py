@@ -220,3 +220,24 @@ class TestTemplatesOnGPU:
         r2 = algo.predict(models[0],
                           Query(user="u1", num=5, black_list=banned))
         assert not set(s.item for s in r2.item_scores) & set(banned)
+
+
+@requires_gpu
+class TestGraphedTopK:
+    def test_graph_matches_eager(self):
+        """hipGraph-captured serving step equals the eager launch and is
+        replayable with new query content."""
+        from predictionio_amd.ops import topk as topk_ops
+        from predictionio_amd.ops.graphs import GraphedTopK
+        g = torch.Generator().manual_seed(21)
+        N, f, K, B = 20000, 64, 10, 4
+        Y = torch.randn((N, f), generator=g).float().cuda()
+        gt = GraphedTopK(Y, K=K, batch=B)
+        for trial in range(3):
+            Xq = torch.randn((B, f), generator=g).float().cuda()
+            gv, gi = gt(Xq)
+            ev, ei = topk_ops.topk_score(Xq, Y, K)
+            assert torch.allclose(gv, ev, atol=1e-4), \
+                f"trial {trial}: {(gv - ev).abs().max()}"
+            chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
+            assert torch.allclose(chosen, ev, atol=1e-4)
