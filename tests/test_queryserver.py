@@ -138,7 +138,11 @@ class TestStopAuth:
             engine_factory=FACTORY, access_key="SECRET")))
         r = c.post("/stop")
         assert r.status_code == 401
-        # correct key accepted (patch os.kill so the test survives)
+        # correct key accepted (patch os.kill so the test survives; the
+        # endpoint fires it from a 0.2s Timer — sleep past it while the
+        # patch is still active, else the real SIGTERM lands mid-suite)
         import os
+        import time
         monkeypatch.setattr(os, "kill", lambda *a: None)
         assert c.post("/stop?accessKey=SECRET").status_code == 200
+        time.sleep(0.4)

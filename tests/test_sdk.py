@@ -1,15 +1,20 @@
-"""SDK client tests — run the real event + engine servers on localhost
-ports (uvicorn in background threads) and drive them with
-predictionio_amd.sdk exactly as an application using the official
-PredictionIO Python SDK would."""
+"""SDK client tests — run the real event + engine servers as subprocesses
+on localhost ports (sharing the sqlite storage file via env) and drive
+them with predictionio_amd.sdk exactly as an application using the
+official PredictionIO Python SDK would."""
 
+import os
 import socket
-import threading
+import subprocess
+import sys
 import time
 
 import pytest
 
 from predictionio_amd.data.storage.base import AccessKey, App
+
+HELPER = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "_sdk_server.py")
 
 
 def _free_port() -> int:
@@ -20,18 +25,32 @@ def _free_port() -> int:
     return p
 
 
-def _serve(app, port):
-    import uvicorn
-    cfg = uvicorn.Config(app, host="127.0.0.1", port=port,
-                         log_level="error")
-    server = uvicorn.Server(cfg)
-    t = threading.Thread(target=server.run, daemon=True)
-    t.start()
-    for _ in range(100):
-        if server.started:
-            return server
-        time.sleep(0.05)
-    raise RuntimeError("server did not start")
+def _wait_port(port: int, proc, timeout=20.0):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if proc.poll() is not None:
+            raise RuntimeError(
+                f"server exited: {proc.stderr.read().decode()[:500]}")
+        try:
+            with socket.create_connection(("127.0.0.1", port), timeout=0.3):
+                return
+        except OSError:
+            time.sleep(0.1)
+    proc.terminate()
+    raise RuntimeError("server did not come up")
+
+
+def _spawn_server(kind: str, *extra) -> tuple:
+    port = _free_port()
+    root = os.path.dirname(os.path.dirname(HELPER))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = root + os.pathsep + env.get("PYTHONPATH", "")
+    proc = subprocess.Popen(
+        [sys.executable, HELPER, kind, str(port), *extra],
+        env=env, stdout=subprocess.DEVNULL, stderr=subprocess.PIPE,
+        cwd=root)
+    _wait_port(port, proc)
+    return proc, f"http://127.0.0.1:{port}"
 
 
 @pytest.fixture()
@@ -40,17 +59,20 @@ def event_server(mem_storage):
     mem_storage.get_meta_data_access_keys().insert(
         AccessKey(key="SDKKEY", appid=app_id, events=[]))
     mem_storage.get_l_events().init(app_id)
-    from predictionio_amd.server.eventserver import create_app
-    port = _free_port()
-    server = _serve(create_app(), port)
-    yield f"http://127.0.0.1:{port}"
-    server.should_exit = True
-    time.sleep(0.1)
+    proc, url = _spawn_server("event")
+    yield url
+    proc.terminate()
+    proc.wait(timeout=10)
 
 
 class TestEventClient:
     def test_lifecycle(self, event_server):
-        from predictionio_amd.sdk import EventClient, NotFoundError
+        from predictionio_amd.sdk import (
+            EventClient, NotCreatedError, NotFoundError,
+        )
+        bad = EventClient("WRONG", event_server)
+        with pytest.raises(NotCreatedError):
+            bad.set_user("u1")
         c = EventClient("SDKKEY", event_server)
         r = c.record_user_action_on_item("rate", "u1", "i1",
                                          {"rating": 4.0})
@@ -65,12 +87,6 @@ class TestEventClient:
         with pytest.raises(NotFoundError):
             c.get_event(eid)
 
-    def test_bad_key_raises(self, event_server):
-        from predictionio_amd.sdk import EventClient, NotCreatedError
-        c = EventClient("WRONG", event_server)
-        with pytest.raises(NotCreatedError):
-            c.set_user("u1")
-
 
 class TestEngineClient:
     def test_query(self, mem_storage):
@@ -80,16 +96,12 @@ class TestEngineClient:
             "tests.fake_engine.JsonEngineFactory",
             "datasource": {"params": {"n": 4}},
             "algorithms": [{"name": "", "params": {}}]})
-        from predictionio_amd.server.queryserver import (
-            ServerConfig, create_app,
-        )
-        port = _free_port()
-        server = _serve(create_app(ServerConfig(
-            engine_factory="tests.fake_engine.JsonEngineFactory")), port)
+        proc, url = _spawn_server(
+            "engine", "tests.fake_engine.JsonEngineFactory")
         try:
             from predictionio_amd.sdk import EngineClient
-            ec = EngineClient(f"http://127.0.0.1:{port}")
+            ec = EngineClient(url)
             assert ec.send_query({"x": 2}) == {"result": 8}
         finally:
-            server.should_exit = True
-            time.sleep(0.1)
+            proc.terminate()
+            proc.wait(timeout=10)
