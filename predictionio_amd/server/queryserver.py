@@ -65,6 +65,7 @@ class ServerConfig:
     access_key: Optional[str] = None
     app_name: Optional[str] = None
     engine_instance_id: Optional[str] = None  # None = latest completed
+    log_url: Optional[str] = None  # remote error log POST target
 
 
 @dataclass
@@ -212,6 +213,20 @@ def create_app(config: ServerConfig,
             prediction = s.serving.serve(query, predictions)
         except Exception as e:
             logger.exception("query failed")
+            if config.log_url:
+                # remote error log (CreateServer.remoteLog :435-446)
+                def _remote_log(msg=str(e)):
+                    import urllib.request
+                    try:
+                        req = urllib.request.Request(
+                            config.log_url,
+                            data=json.dumps({"level": "ERROR",
+                                             "message": msg}).encode(),
+                            headers={"Content-Type": "application/json"})
+                        urllib.request.urlopen(req, timeout=5)
+                    except Exception:
+                        logger.exception("remote log POST failed")
+                threading.Thread(target=_remote_log, daemon=True).start()
             return JSONResponse({"message": str(e)}, status_code=500)
         for b in blockers:
             try:
