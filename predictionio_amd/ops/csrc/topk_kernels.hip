@@ -138,24 +138,40 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     }
     __syncthreads();
 
-    // wave w scores items [16w, 16w+16) of the chunk for all 64 users
+    // wave w scores items [8w, 8w+8) of the chunk for all 64 users, in
+    // GROUPS OF 4: serve PMC showed ~370 wait cycles per item — the
+    // per-item threshold branch serialized the next item's LDS reads
+    // behind the current item's score. Four interleaved dot chains keep
+    // 4 items' reads in flight; the (rare) mask/ban checks move into the
+    // insert path.
     const int c_lo = wave * (TK_CHUNK / TK_WAVES);
     const int c_hi = min(cn, c_lo + TK_CHUNK / TK_WAVES);
-    for (int c = c_lo; c < c_hi; ++c) {
-      const long long item = base + c;
-      if (item_mask != nullptr && item_mask[item]) continue;
-      const f32x2* yrow = reinterpret_cast<const f32x2*>(ys + c * FP);
+    // GRP=2 fits the 12-waves/CU register budget at F<=64; F=128's
+    // xreg alone is 128 VGPRs, so it stays at one chain.
+    constexpr int GRP = F <= 64 ? 2 : 1;
+    for (int c0 = c_lo; c0 < c_hi; c0 += GRP) {
+      const int gn = min(GRP, c_hi - c0);
       f32x2 a0 = {0.f, 0.f}, a1 = {0.f, 0.f};
+      const f32x2* yr0 = reinterpret_cast<const f32x2*>(ys + (c0 + 0) * FP);
+      const f32x2* yr1 = reinterpret_cast<const f32x2*>(
+          ys + (c0 + (GRP > 1 ? 1 : 0)) * FP);
+      // FULLY unrolled: q must be a literal so xreg[] stays in registers
+      // (a capped unroll sent it to scratch).
 #pragma unroll
-      for (int q = 0; q + 1 < F / 2; q += 2) {
-        a0 += xreg[q] * yrow[q];       // v_pk_fma_f32, 2 indep chains
-        a1 += xreg[q + 1] * yrow[q + 1];
+      for (int q = 0; q < F / 2; ++q) {
+        a0 += xreg[q] * yr0[q];   // v_pk_fma_f32, indep item chains
+        if (GRP > 1) a1 += xreg[q] * yr1[q];
       }
-      if (F / 2 & 1) a0 += xreg[F / 2 - 1] * yrow[F / 2 - 1];
-      const float s = a0.x + a0.y + a1.x + a1.y;
+      float sg[2] = {a0.x + a0.y, a1.x + a1.y};
+#pragma unroll
+      for (int g = 0; g < GRP; ++g) {
+      const int c = c0 + g;
+      const long long item = base + c;
+      const float s = g < gn ? sg[g] : -FLT_MAX;
       if (has_user && s > th) {
-        if (ban == nullptr ||
-            !in_sorted(ban, bn, (int)(item + item_base))) {
+        if ((item_mask == nullptr || !item_mask[item]) &&
+            (ban == nullptr ||
+             !in_sorted(ban, bn, (int)(item + item_base)))) {
           // replace current min of this lane's K-list
           int mi = 0;
           float mv = tvu[0];
@@ -167,6 +183,7 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
           for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
           th = nm;
         }
+      }
       }
     }
   }
