@@ -99,6 +99,10 @@ def run_train(args, device):
     elapsed = time.time() - t0
     elapsed = pdist.max_scalar(elapsed)  # slowest rank
     sec_per_iter = elapsed / args.steps
+    if os.environ.get("PIO_PHASE_TIMES") == "1":
+        pt = trainer.phase_times
+        log(f"phase times over timed steps: gather {pt['gather_s']:.3f}s "
+            f"solve {pt['solve_s']:.3f}s")
     total_nnz = args.users_per_gpu * args.nnz_per_user * world
     value = total_nnz / sec_per_iter  # ratings solved per second per iter
     return value, sec_per_iter, {

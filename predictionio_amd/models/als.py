@@ -17,6 +17,8 @@ hyperparameters — on an MI355X-native substrate:
 
 from __future__ import annotations
 
+import os
+import time
 from dataclasses import dataclass
 from typing import Optional, Tuple
 
@@ -59,6 +61,7 @@ class ALSTrainer:
         self.item_csr = None   # (indptr, indices, values) local items x users
         self.X: Optional[torch.Tensor] = None  # local user factors
         self.Y: Optional[torch.Tensor] = None  # local item factors
+        self.phase_times = {"gather_s": 0.0, "solve_s": 0.0}
 
     # ------------------------------------------------------------ data
 
@@ -110,14 +113,34 @@ class ALSTrainer:
 
     def _half_step(self, csr, fixed_local: torch.Tensor,
                    n_fixed: int) -> torch.Tensor:
-        """One half-iteration: all-gather the fixed side, fused solve."""
+        """One half-iteration: all-gather the fixed side, fused solve.
+
+        Set PIO_PHASE_TIMES=1 to accumulate per-phase wall times
+        (gather / solve) into self.phase_times — the diagnostic for
+        reading multi-GPU scaling results."""
+        timing = os.environ.get("PIO_PHASE_TIMES") == "1"
+        if timing:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t0 = time.time()
         fixed_full = pdist.all_gather_rows(fixed_local, n_fixed)
+        if timing:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t1 = time.time()
         yty = als_ops.gramian(fixed_full) if self.p.implicit else None
         indptr, indices, values = csr
-        return als_ops.als_solve(
+        out = als_ops.als_solve(
             indptr, indices, values, fixed_full, YtY=yty,
             lam=self.p.lambda_, alpha=self.p.alpha,
             implicit=self.p.implicit, wr_scale=not self.p.implicit)
+        if timing:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t2 = time.time()
+            self.phase_times["gather_s"] += t1 - t0
+            self.phase_times["solve_s"] += t2 - t1
+        return out
 
     def step(self) -> None:
         """One full ALS iteration (user half-step then item half-step)."""
