@@ -92,6 +92,7 @@ def run_train(args, device):
     for w in range(args.warmup):
         trainer.step()
     sync()
+    trainer.phase_times = {"gather_s": 0.0, "solve_s": 0.0}
     t0 = time.time()
     for k in range(args.steps):
         trainer.step()
