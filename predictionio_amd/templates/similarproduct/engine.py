@@ -311,3 +311,64 @@ class SimilarProductEngine(EngineFactory):
                              "cooccurrence": CooccurrenceAlgorithm,
                              "": ALSAlgorithm},
             serving_class=Serving)
+
+
+# --------------------------------------------------------------------------
+# recommended-user variant (examples/scala-parallel-similarproduct/
+# recommended-user): the same cosine-kNN machinery over USER factors —
+# "users who follow the query users also follow these users". Events are
+# `follow` (user → similarUser); the model keeps the followed-user factors.
+# --------------------------------------------------------------------------
+
+
+@dataclass
+class UserQuery:
+    users: List[str]
+    num: int
+    white_list: Optional[List[str]] = None
+    black_list: Optional[List[str]] = None
+
+    @staticmethod
+    def from_json(d: Dict[str, Any]) -> "UserQuery":
+        return UserQuery(users=list(d["users"]), num=int(d.get("num", 10)),
+                         white_list=d.get("whiteList"),
+                         black_list=d.get("blackList"))
+
+
+class FollowDataSource(BaseDataSource):
+    """Reads `follow` events (user → similarUser entity type `user`)."""
+
+    def read_training(self) -> TrainingData:
+        app = self.params["appName"]
+        follows = [
+            ViewEvent(e.entity_id, e.target_entity_id,
+                      e.event_time.timestamp())
+            for e in event_store.find(app, entity_type="user",
+                                      event_names=["follow"],
+                                      target_entity_type="user")
+        ]
+        return TrainingData({}, {}, follows)
+
+
+class RecommendedUserAlgorithm(ALSAlgorithm):
+    """Trains implicit ALS on follow events; the 'item' side is the
+    followed users, so predict returns similar USERS
+    (recommended-user ALSAlgorithm.scala)."""
+
+    def predict(self, model: SimilarModel, query) -> PredictedResult:
+        q = query if isinstance(query, UserQuery) \
+            else UserQuery.from_json(query)
+        return super().predict(
+            model, Query(items=q.users, num=q.num,
+                         white_list=q.white_list, black_list=q.black_list))
+
+
+class RecommendedUserEngine(EngineFactory):
+    @classmethod
+    def apply(cls) -> Engine:
+        return Engine(
+            data_source_class=FollowDataSource,
+            preparator_class=Preparator,
+            algorithm_class={"als": RecommendedUserAlgorithm,
+                             "": RecommendedUserAlgorithm},
+            serving_class=Serving)

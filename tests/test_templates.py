@@ -359,3 +359,27 @@ class TestRandomForestVariant:
         algo = ce.RandomForestAlgorithm(ep.algorithms_params[0][1])
         assert algo.predict(models[0], ce.Query(9, 5, 1)).label == 0.0
         assert algo.predict(models[0], ce.Query(1, 5, 9)).label == 1.0
+
+
+class TestRecommendedUserVariant:
+    def test_similar_users(self, mem_storage):
+        app_id = _mk_app(mem_storage, "FollowApp")
+        # two follow communities: even users follow even targets
+        for u in range(30):
+            for t in (u % 2, u % 2 + 2, u % 2 + 4):
+                _insert(mem_storage, app_id, "follow", f"u{u}", f"s{t}",
+                        target_type="user")
+        from predictionio_amd.templates.similarproduct.engine import (
+            RecommendedUserAlgorithm, RecommendedUserEngine, UserQuery,
+        )
+        from predictionio_amd.controller import EngineParams, Params
+        e = RecommendedUserEngine.apply()
+        ep = EngineParams(
+            data_source_params=Params({"appName": "FollowApp"}),
+            algorithms_params=[("als", Params(
+                {"rank": 8, "numIterations": 8, "seed": 4}))])
+        models = e.train(ep)
+        algo = RecommendedUserAlgorithm(ep.algorithms_params[0][1])
+        r = algo.predict(models[0], UserQuery(users=["s0"], num=3))
+        assert len(r.item_scores) == 3
+        assert "s0" not in [s.item for s in r.item_scores]
