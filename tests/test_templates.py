@@ -383,3 +383,56 @@ class TestRecommendedUserVariant:
         r = algo.predict(models[0], UserQuery(users=["s0"], num=3))
         assert len(r.item_scores) == 3
         assert "s0" not in [s.item for s in r.item_scores]
+
+
+class TestMultiAlgoEngine:
+    def test_als_plus_cooccurrence_combined(self, mem_storage):
+        """multi-events-multi-algos variant shape: one engine instance
+        running BOTH algorithms, serving combining their predictions."""
+        app_id = _mk_app(mem_storage)
+        TestSimilarProductTemplate._seed(TestSimilarProductTemplate(),
+                                         mem_storage, app_id)
+        from predictionio_amd.controller import (
+            Engine, EngineParams, Params, Serving,
+        )
+        from predictionio_amd.templates.similarproduct.engine import (
+            ALSAlgorithm, CooccurrenceAlgorithm, DataSource, ItemScore,
+            PredictedResult, Preparator, Query,
+        )
+
+        class CombinedServing(Serving):
+            """Interleave algorithm outputs, dedup by item (the variant's
+            custom Serving combines per-algorithm scores)."""
+
+            def serve(self, q, preds):
+                seen, out = set(), []
+                for rank_pos in range(max(len(p.item_scores)
+                                          for p in preds)):
+                    for p in preds:
+                        if rank_pos < len(p.item_scores):
+                            s = p.item_scores[rank_pos]
+                            if s.item not in seen:
+                                seen.add(s.item)
+                                out.append(s)
+                return PredictedResult(out[:q.num])
+
+        e = Engine(DataSource, Preparator,
+                   {"als": ALSAlgorithm,
+                    "cooccurrence": CooccurrenceAlgorithm},
+                   CombinedServing)
+        ep = EngineParams(
+            data_source_params=Params({"appName": "MyTestApp"}),
+            algorithms_params=[
+                ("als", Params({"rank": 8, "numIterations": 5,
+                                "seed": 7})),
+                ("cooccurrence", Params({"n": 10}))])
+        models = e.train(ep)
+        assert len(models) == 2
+        algos = e._algorithms(ep)
+        serving = e._serving(ep)
+        q = Query(items=["i0"], num=5)
+        preds = [a.predict(m, q) for a, m in zip(algos, models)]
+        combined = serving.serve(q, preds)
+        assert 1 <= len(combined.item_scores) <= 5
+        assert len({s.item for s in combined.item_scores}) == \
+            len(combined.item_scores)  # deduped
