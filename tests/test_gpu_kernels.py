@@ -241,3 +241,46 @@ class TestGraphedTopK:
                 f"trial {trial}: {(gv - ev).abs().max()}"
             chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
             assert torch.allclose(chosen, ev, atol=1e-4)
+
+
+@requires_gpu
+class TestEvalOnGPU:
+    def test_precision_at_k_eval(self, mem_storage):
+        """k-fold Precision@K evaluation with device-batched predictions
+        (the reference's eval hot loop, Engine.scala:771-786, with
+        batchPredict = one fused top-K launch per fold)."""
+        import random
+        from predictionio_amd.data.events import DataMap, Event, utcnow
+        from predictionio_amd.data.storage.base import App
+        app_id = mem_storage.get_meta_data_apps().insert(
+            App(id=0, name="EvalApp"))
+        le = mem_storage.get_l_events()
+        le.init(app_id)
+        rng = random.Random(9)
+        for u in range(40):
+            for i in rng.sample([i for i in range(24)
+                                 if i % 2 == u % 2], 6):
+                le.insert(Event(
+                    event="rate", entity_type="user", entity_id=f"u{u}",
+                    target_entity_type="item", target_entity_id=f"i{i}",
+                    properties=DataMap({"rating": rng.uniform(3.5, 5)}),
+                    event_time=utcnow()), app_id)
+        from predictionio_amd.controller import (
+            EngineParams, MetricEvaluator, Params,
+        )
+        from predictionio_amd.templates.recommendation import (
+            RecommendationEngine,
+        )
+        from predictionio_amd.templates.recommendation.evaluation import (
+            PrecisionAtK,
+        )
+        e = RecommendationEngine.apply()
+        ep = EngineParams(
+            data_source_params=Params({"appName": "EvalApp",
+                                       "evalParams": {"kFold": 2,
+                                                      "queryNum": 5}}),
+            algorithms_params=[("als", Params(
+                {"rank": 16, "numIterations": 5, "seed": 2}))])
+        res = MetricEvaluator(PrecisionAtK(k=5)).evaluate_base(
+            e, e.batch_eval([ep]))
+        assert 0.0 <= res.best_score <= 1.0
