@@ -1,8 +1,11 @@
 #!/bin/bash
 # Round-2 starter: advanced thread trace of the Woodbury kernel (SQ
 # counters cannot localize its remaining per-row latency — NOTES.md).
-# UNTESTED in round 1 (left as the prepared next step). Keep --pmc OUT of
-# this invocation (gpurun refuses pmc combined with trace domains).
+# TESTED round 1: FAILS on this image — rocprofv3 aborts with
+# "rocprof-trace-decoder library path not found" (decoder not shipped in
+# ROCm 7.2.0 here). Round-2 fallback: instrument the kernel itself with
+# s_memtime deltas written to a debug buffer per phase (stage/G/solve),
+# which needs no tooling. Keep --pmc OUT of any trace invocation.
 set -x
 export TMPDIR=/tmp
 cd /tmp
