@@ -431,7 +431,12 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     float lambda,
     float alpha,
     int implicit_mode,
-    int wr_scale)
+    int wr_scale,
+    unsigned long long* prof)        // optional [5]: phase wall-clock sums
+                                     // (stage, G, solve, emit, samples) —
+                                     // ATT is unavailable on this image,
+                                     // so the kernel self-times sampled
+                                     // rows with wall_clock64()
 {
   constexpr int NW = WOODBURY_MAX_NNZ;
   constexpr int FP = F + 4;  // bank-group padding
@@ -457,6 +462,9 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
         X[row * (long long)F + e] = 0.f;
       continue;
     }
+    const bool probe = prof != nullptr && (row & 1023) == 0 && lane == 0;
+    unsigned long long pt0 = 0, pt1 = 0, pt2 = 0, pt3 = 0;
+    if (probe) pt0 = wall_clock64();
 
     // ---- stage factor rows + per-item weights ----
     // One coalesced load of the column ids first, THEN the factor-row
@@ -487,6 +495,7 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
       }
     }
     wave_sync();
+    if (probe) pt1 = wall_clock64();
 
     // ---- M = I + D G D  (implicit)  or  G + reg I  (explicit) ----
     // G is symmetric (G_ij = y_i^T B^-1 y_j): compute the upper triangle
@@ -521,6 +530,7 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     }
     wave_sync();
 
+    if (probe) pt2 = wall_clock64();
     // ---- Cholesky of M: register rows + quad pivot-column reads ----
     // lane = row of M in registers (mr); each finished column is written
     // into M's row k (M[k][j] = L[j][k]) and the trailing update reads it
@@ -591,6 +601,7 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     if (lane < n) tv[wave][lane] = t;
     wave_sync();
 
+    if (probe) pt3 = wall_clock64();
     // ---- emit sum_i s_i v_i (implicit: z, host solves X = Z L^-1)
     //      or   sum_i s_i y_i (explicit: x directly) ----
     for (int e = lane; e < F; e += 64) {
@@ -598,6 +609,14 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
       for (int c = 0; c < n; ++c)
         x = fmaf(tv[wave][c], yl[c * FP + e], x);
       X[row * (long long)F + e] = x;
+    }
+    if (probe) {
+      const unsigned long long pt4 = wall_clock64();
+      atomicAdd(&prof[0], pt1 - pt0);   // stage
+      atomicAdd(&prof[1], pt2 - pt1);   // G build
+      atomicAdd(&prof[2], pt3 - pt2);   // M solve
+      atomicAdd(&prof[3], pt4 - pt3);   // emit
+      atomicAdd(&prof[4], 1ull);        // samples
     }
     wave_sync();
   }
@@ -613,7 +632,8 @@ extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
     const float* Y, const float* YtY, const float* V, float* X,
     int n_rows, int f, float lambda, float alpha,
-    int implicit_mode, int wr_scale, int which, hipStream_t stream)
+    int implicit_mode, int wr_scale, int which,
+    unsigned long long* prof, hipStream_t stream)
 {
   // >> 256 workgroups to fill 256 CUs across 8 XCDs; grid-stride for huge
   // row counts.
@@ -644,7 +664,7 @@ extern "C" void launch_als_solve(
   if (woodbury)                                                              \
   hipLaunchKernelGGL((als_woodbury_kernel<FF>), dim3(grid_w), dim3(128),     \
                      0, stream, indptr, indices, values, Y, V, X, n_rows,    \
-                     lambda, alpha, implicit_mode, wr_scale)
+                     lambda, alpha, implicit_mode, wr_scale, prof)
   switch (f) {
     case 16: LAUNCH_WOODBURY(16); LAUNCH_WAVE(16); break;
     case 32: LAUNCH_WOODBURY(32); LAUNCH_WAVE(32); break;

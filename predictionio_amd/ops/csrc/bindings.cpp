@@ -14,7 +14,8 @@ extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
     const float* Y, const float* YtY, const float* V, float* X,
     int n_rows, int f, float lambda, float alpha,
-    int implicit_mode, int wr_scale, int which, hipStream_t stream);
+    int implicit_mode, int wr_scale, int which,
+    unsigned long long* prof, hipStream_t stream);
 
 extern "C" void launch_topk_score(
     const float* Xq, const float* Y, const uint8_t* item_mask,
@@ -49,7 +50,8 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
                         c10::optional<torch::Tensor> V,
                         double lambda, double alpha,
                         bool implicit_mode, bool wr_scale,
-                        int64_t which, c10::optional<torch::Tensor> out) {
+                        int64_t which, c10::optional<torch::Tensor> out,
+                        c10::optional<torch::Tensor> prof) {
   TORCH_CHECK(indptr.is_cuda() && indptr.scalar_type() == torch::kInt64 &&
                   indptr.is_contiguous(), "indptr must be contiguous i64 GPU");
   TORCH_CHECK(indices.is_cuda() && indices.scalar_type() == torch::kInt32 &&
@@ -72,6 +74,15 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
     TORCH_CHECK(V->sizes() == Y.sizes(), "V must match Y shape");
     v_ptr = V->data_ptr<float>();
   }
+  unsigned long long* prof_ptr = nullptr;
+  if (prof.has_value()) {
+    TORCH_CHECK(prof->is_cuda() && prof->is_contiguous() &&
+                prof->scalar_type() == torch::kUInt64 &&
+                prof->numel() >= 5,
+                "prof must be a contiguous u64 GPU tensor of >= 5");
+    prof_ptr = reinterpret_cast<unsigned long long*>(
+        prof->data_ptr());
+  }
   torch::Tensor X;
   if (out.has_value()) {
     check_cuda_f32(*out, "out");
@@ -87,7 +98,7 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
                    values.data_ptr<float>(), Y.data_ptr<float>(), yty_ptr,
                    v_ptr, X.data_ptr<float>(), (int)n_rows, (int)f, (float)lambda,
                    (float)alpha, implicit_mode ? 1 : 0, wr_scale ? 1 : 0,
-                   (int)which, stream);
+                   (int)which, prof_ptr, stream);
   C10_HIP_CHECK(hipGetLastError());
   return X;
 }
@@ -152,7 +163,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("lambda_") = 0.01,
         py::arg("alpha") = 1.0, py::arg("implicit_mode") = false,
         py::arg("wr_scale") = true, py::arg("which") = 0,
-        py::arg("out") = py::none());
+        py::arg("out") = py::none(), py::arg("prof") = py::none());
   m.def("topk_score", &topk_score, "Fused masked top-K scoring",
         py::arg("Xq"), py::arg("Y"), py::arg("K"), py::arg("n_slices") = 64,
         py::arg("item_mask") = py::none(), py::arg("ban_indptr") = py::none(),
