@@ -162,3 +162,91 @@ class TestFsspecModels:
         assert models.delete("abc") is True
         assert models.get("abc") is None
         storage.reset()
+
+
+class TestServerProcesses:
+    """Launch the real `pio eventserver` / `pio deploy` commands as
+    processes (the reference integration harness drives the CLI the same
+    way, tests/pio_tests/utils.py AppEngine.deploy)."""
+
+    def _spawn(self, args, env):
+        import subprocess, sys, os, socket, time
+        root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env = dict(env)
+        env["PYTHONPATH"] = root + os.pathsep + env.get("PYTHONPATH", "")
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "predictionio_amd.cli.main", *args],
+            env=env, cwd=root, stdout=subprocess.DEVNULL,
+            stderr=subprocess.PIPE)
+        return proc
+
+    def _wait(self, proc, port, timeout=25):
+        import socket, time
+        t0 = time.time()
+        while time.time() - t0 < timeout:
+            if proc.poll() is not None:
+                raise RuntimeError(proc.stderr.read().decode()[:400])
+            try:
+                with socket.create_connection(("127.0.0.1", port),
+                                              timeout=0.3):
+                    return
+            except OSError:
+                time.sleep(0.15)
+        proc.terminate()
+        raise RuntimeError("no listen")
+
+    def test_pio_eventserver_and_deploy(self, mem_storage, tmp_path):
+        import json as js
+        import os
+        import socket
+        import urllib.request
+        from predictionio_amd.data.storage.base import AccessKey, App
+        app_id = mem_storage.get_meta_data_apps().insert(App(0, "ProcApp"))
+        mem_storage.get_meta_data_access_keys().insert(
+            AccessKey(key="PKEY", appid=app_id, events=[]))
+        mem_storage.get_l_events().init(app_id)
+
+        def free_port():
+            s = socket.socket(); s.bind(("127.0.0.1", 0))
+            p = s.getsockname()[1]; s.close(); return p
+
+        ep_port = free_port()
+        es = self._spawn(["eventserver", "--ip", "127.0.0.1",
+                          "--port", str(ep_port)], os.environ)
+        try:
+            self._wait(es, ep_port)
+            body = js.dumps({"event": "rate", "entityType": "user",
+                             "entityId": "u1", "targetEntityType": "item",
+                             "targetEntityId": "i1",
+                             "properties": {"rating": 5}}).encode()
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{ep_port}/events.json?accessKey=PKEY",
+                data=body, headers={"Content-Type": "application/json"})
+            assert js.loads(urllib.request.urlopen(req, timeout=5).read()
+                            )["eventId"]
+        finally:
+            es.terminate(); es.wait(timeout=10)
+
+        # train a JSON engine then `pio deploy` it as a process
+        from predictionio_amd.workflow import train as train_wf
+        variant = {"id": "proc",
+                   "engineFactory": "tests.fake_engine.JsonEngineFactory",
+                   "datasource": {"params": {"n": 4}},
+                   "algorithms": [{"name": "", "params": {}}]}
+        train_wf.run_train_from_variant(variant)
+        vpath = tmp_path / "engine.json"
+        vpath.write_text(js.dumps(variant))
+        qs_port = free_port()
+        qs = self._spawn(["deploy", "--engine-dir", str(tmp_path),
+                          "--ip", "127.0.0.1", "--port", str(qs_port)],
+                         os.environ)
+        try:
+            self._wait(qs, qs_port)
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{qs_port}/queries.json",
+                data=js.dumps({"x": 3}).encode(),
+                headers={"Content-Type": "application/json"})
+            assert js.loads(urllib.request.urlopen(req, timeout=5).read()
+                            ) == {"result": 9}
+        finally:
+            qs.terminate(); qs.wait(timeout=10)
