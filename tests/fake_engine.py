@@ -169,3 +169,12 @@ def make_params(algos=None, **kw) -> EngineParams:
         algorithms_params=algos or [("algo0", Params())],
         serving_params=Params(kw.get("serving", {})),
     )
+
+
+RUN_CALLS = []
+
+
+def fake_main(*args):
+    """Target for `pio run` tests (FakeWorkflow.FakeRun parity)."""
+    RUN_CALLS.append(args)
+    return len(args)

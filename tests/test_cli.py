@@ -112,3 +112,15 @@ class TestStatus:
         r = runner.invoke(cli, ["status"])
         assert r.exit_code == 0, r.output
         assert "ready to go" in r.output
+
+
+class TestRunCommand:
+    def test_run_arbitrary_main(self, runner):
+        """`pio run pkg.mod.fn args...` (commands/Engine.run + FakeWorkflow
+        parity: run an arbitrary entry point under the PIO environment)."""
+        from tests import fake_engine
+        fake_engine.RUN_CALLS.clear()
+        r = runner.invoke(cli, ["run", "tests.fake_engine.fake_main",
+                                "a", "b"])
+        assert r.exit_code == 0, r.output
+        assert fake_engine.RUN_CALLS == [("a", "b")]
