@@ -53,4 +53,9 @@ def main(n_rows=2_000_000, n_cols=1_000_000, f=64, npr=20):
 
 
 if __name__ == "__main__":
-    main()
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--npr", type=int, default=20)
+    ap.add_argument("--rows", type=int, default=2_000_000)
+    a = ap.parse_args()
+    main(n_rows=a.rows, npr=a.npr)
