@@ -665,7 +665,11 @@ extern "C" void launch_als_solve(
                      lambda, alpha, implicit_mode, wr_scale, skip)
 #define LAUNCH_WOODBURY(FF)                                                  \
   if (woodbury) {                                                            \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 24, -1>), dim3(grid_w),      \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 20, -1>), dim3(grid_w),      \
+                       dim3(128), 0, stream, indptr, indices, values, Y, V,  \
+                       X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
+                       prof);                                                \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 24, 20>), dim3(grid_w),      \
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \
                        X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
                        prof);                                                \
