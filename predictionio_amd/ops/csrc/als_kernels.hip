@@ -26,6 +26,13 @@
 
 #define CHUNK 16  // rated rows staged per LDS pass
 
+__device__ __forceinline__ void wave_sync() {
+  __builtin_amdgcn_wave_barrier();
+}
+
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+
 template <int F>
 __global__ __launch_bounds__(256) void als_solve_kernel(
     const long long* __restrict__ indptr,   // n_rows+1
@@ -44,18 +51,25 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
   constexpr int TM = 4;                       // thread tile edge
   constexpr int TILES = (F / TM) * (F / TM);  // tiles covering FxF
   constexpr int TPT = (TILES + 255) / 256;    // tiles per thread
+  constexpr int NB = 16;                      // Cholesky panel width
   static_assert(F % TM == 0, "F must be divisible by 4");
+  static_assert(F % NB == 0, "F must be divisible by the panel width");
 
   // +1 padding on the staging buffer breaks the F-stride bank conflict
   // (MI355X LDS = 32 banks x 4 B; F=64 floats stride = same-bank).
   __shared__ float ys[CHUNK][F + 1];
   __shared__ float ws_a[CHUNK];  // Gramian weight per staged row
   __shared__ float ws_b[CHUNK];  // b-vector weight per staged row
-  __shared__ float As[F][F + 1];
+  // As rows padded to F+4: row starts stay 16-byte aligned, so the
+  // rank-NB trailing update can read L panels as float4 quads (the same
+  // ds_read_b128 lesson as the wave kernel's pivot columns).
+  __shared__ float As[F][F + 4];
   __shared__ float bs[F];
-  __shared__ float diag_inv;
+  __shared__ float dinv_all[F];  // 1/L[k][k], filled during factorization
 
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
 
   for (long long row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const long long start = indptr[row];
@@ -154,43 +168,144 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
     }
     __syncthreads();
 
-    // ---- in-LDS Cholesky (right-looking, lower triangle) ----
-    for (int k = 0; k < F; ++k) {
-      if (tid == 0) {
-        float d = As[k][k];
-        d = d > 0.f ? sqrtf(d) : 1e-20f;
-        As[k][k] = d;
-        diag_inv = 1.f / d;
+    // ---- in-LDS blocked Cholesky (NB-wide panels, lower triangle) ----
+    // The unblocked right-looking loop ran 3 __syncthreads per pivot
+    // (3F = 384 s_barriers at F=128) and measured ~450 us/row of nearly
+    // pure barrier/serialization cost (microbench: time flat in nnz).
+    // Blocked form: per panel, (1) wave 0 factors the NBxNB diagonal
+    // block alone — intra-wave lockstep + wave_sync() instead of
+    // s_barrier, the same free-running idiom as the wave kernel above;
+    // (2) the panel below it is a triangular solve with INDEPENDENT
+    // rows — one row per thread, no barriers; (3) a rank-NB trailing
+    // update amortizes one barrier over NB pivots. 3 s_barriers per
+    // panel = 24 at F=128 instead of 384.
+    for (int k0 = 0; k0 < F; k0 += NB) {
+      if (wave == 0) {
+        // (1) diagonal block, lanes 0..NB-1 of wave 0 (lane i = row i)
+        for (int kk = 0; kk < NB; ++kk) {
+          const int kq = k0 + kk;
+          wave_sync();
+          float d = As[kq][kq];
+          d = d > 0.f ? sqrtf(d) : 1e-20f;
+          const float dinv = 1.f / d;     // redundant on all lanes
+          if (lane == kk) {
+            As[kq][kq] = d;
+            dinv_all[kq] = dinv;
+          }
+          if (lane > kk && lane < NB) As[k0 + lane][kq] *= dinv;
+          wave_sync();
+          if (lane > kk && lane < NB) {
+            const float lik = As[k0 + lane][kq];
+            for (int j = kk + 1; j <= lane; ++j)
+              As[k0 + lane][k0 + j] =
+                  fmaf(-lik, As[k0 + j][kq], As[k0 + lane][k0 + j]);
+          }
+        }
       }
       __syncthreads();
-      const float dinv = diag_inv;
-      for (int j = k + 1 + tid; j < F; j += 256) As[j][k] *= dinv;
+      if (k0 + NB >= F) break;  // no panel below the last block
+      // (2) panel solve: row i of A[k0+NB.., k0..k0+NB) <- row * L_bb^-T.
+      // Rows are independent: one thread per row, registers w[NB]
+      // literal-indexed via full unroll (runtime indices would spill).
+      for (int i = k0 + NB + tid; i < F; i += 256) {
+        float w[NB];
+#pragma unroll
+        for (int m = 0; m < NB; ++m) w[m] = As[i][k0 + m];
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+          float s = w[j];
+#pragma unroll
+          for (int m = 0; m < NB; ++m)
+            if (m < j) s = fmaf(-w[m], As[k0 + j][k0 + m], s);
+          w[j] = s * dinv_all[k0 + j];
+        }
+#pragma unroll
+        for (int m = 0; m < NB; ++m) As[i][k0 + m] = w[m];
+      }
       __syncthreads();
-      // trailing update: A[i][j] -= L[i][k] * L[j][k] for i>=j>k
-      const int rem = F - k - 1;
-      for (int e = tid; e < rem * rem; e += 256) {
-        const int i = k + 1 + e / rem;
-        const int j = k + 1 + e % rem;
-        if (j <= i) As[i][j] = fmaf(-As[i][k], As[j][k], As[i][j]);
+      // (3) rank-NB trailing update: A[i][j] -= sum_m L[i][m] L[j][m].
+      // 2x2 tiles, panel rows read as float4 quads (rows are 16B-aligned
+      // and k0 is a multiple of NB=16): 4 ds_read_b128 per 4 m-terms of
+      // 4 elements instead of 32 scalar ds_read_b32.
+      const int rem = F - k0 - NB;
+      const int remt = rem >> 1;
+      for (int e = tid; e < remt * remt; e += 256) {
+        const int i = k0 + NB + 2 * (e / remt);
+        const int j = k0 + NB + 2 * (e % remt);
+        if (j <= i) {
+          float s00 = As[i][j];
+          float s01 = As[i][j + 1];
+          float s10 = As[i + 1][j];
+          float s11 = As[i + 1][j + 1];
+#pragma unroll
+          for (int m = 0; m < NB; m += 4) {
+            const f32x4_t a0 =
+                *reinterpret_cast<const f32x4_t*>(&As[i][k0 + m]);
+            const f32x4_t a1 =
+                *reinterpret_cast<const f32x4_t*>(&As[i + 1][k0 + m]);
+            const f32x4_t b0 =
+                *reinterpret_cast<const f32x4_t*>(&As[j][k0 + m]);
+            const f32x4_t b1 =
+                *reinterpret_cast<const f32x4_t*>(&As[j + 1][k0 + m]);
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+              s00 = fmaf(-a0[q], b0[q], s00);
+              s01 = fmaf(-a0[q], b1[q], s01);
+              s10 = fmaf(-a1[q], b0[q], s10);
+              s11 = fmaf(-a1[q], b1[q], s11);
+            }
+          }
+          As[i][j] = s00;
+          if (j + 1 <= i) As[i][j + 1] = s01;
+          As[i + 1][j] = s10;
+          As[i + 1][j + 1] = s11;
+        }
       }
       __syncthreads();
     }
 
-    // ---- forward solve L z = b (z stored back into bs) ----
-    for (int k = 0; k < F; ++k) {
-      if (tid == 0) bs[k] /= As[k][k];
+    // ---- forward solve L z = b, blocked the same way ----
+    for (int k0 = 0; k0 < F; k0 += NB) {
+      if (wave == 0) {
+        for (int kk = 0; kk < NB; ++kk) {
+          wave_sync();
+          const float zk = bs[k0 + kk] * dinv_all[k0 + kk];
+          if (lane == kk) bs[k0 + kk] = zk;
+          if (lane > kk && lane < NB)
+            bs[k0 + lane] = fmaf(-As[k0 + lane][k0 + kk], zk,
+                                 bs[k0 + lane]);
+        }
+      }
       __syncthreads();
-      const float zk = bs[k];
-      for (int j = k + 1 + tid; j < F; j += 256)
-        bs[j] = fmaf(-As[j][k], zk, bs[j]);
+      for (int j = k0 + NB + tid; j < F; j += 256) {
+        float s = bs[j];
+#pragma unroll
+        for (int m = 0; m < NB; ++m)
+          s = fmaf(-As[j][k0 + m], bs[k0 + m], s);
+        bs[j] = s;
+      }
       __syncthreads();
     }
     // ---- back solve L^T x = z ----
-    for (int k = F - 1; k >= 0; --k) {
-      if (tid == 0) bs[k] /= As[k][k];
+    for (int k0 = F - NB; k0 >= 0; k0 -= NB) {
+      if (wave == 0) {
+        for (int kk = NB - 1; kk >= 0; --kk) {
+          wave_sync();
+          const float xk = bs[k0 + kk] * dinv_all[k0 + kk];
+          if (lane == kk) bs[k0 + kk] = xk;
+          if (lane < kk)
+            bs[k0 + lane] = fmaf(-As[k0 + kk][k0 + lane], xk,
+                                 bs[k0 + lane]);
+        }
+      }
       __syncthreads();
-      const float xk = bs[k];
-      for (int j = tid; j < k; j += 256) bs[j] = fmaf(-As[k][j], xk, bs[j]);
+      for (int j = tid; j < k0; j += 256) {
+        float s = bs[j];
+#pragma unroll
+        for (int m = 0; m < NB; ++m)
+          s = fmaf(-As[k0 + m][j], bs[k0 + m], s);
+        bs[j] = s;
+      }
       __syncthreads();
     }
 
@@ -224,10 +339,6 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
 //     row from broadcast float4 reads; the next item's global load is
 //     issued before the current item's math (double-buffered LDS).
 // ---------------------------------------------------------------------------
-
-__device__ __forceinline__ void wave_sync() {
-  __builtin_amdgcn_wave_barrier();
-}
 
 template <int F>
 __global__ __launch_bounds__(128) void als_solve_wave_kernel(
@@ -417,7 +528,6 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
 
 #define WOODBURY_MAX_NNZ 32
 
-typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 // Two NW instantiations: rows with nnz <= 24 run the NW=24 variant whose
 // smaller LDS footprint (9.7 vs 13.3 KB/wave) fits 16 waves/CU instead of

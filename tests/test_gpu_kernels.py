@@ -57,7 +57,7 @@ class TestALSKernel:
         assert torch.allclose(X, X_ref, atol=2e-3, rtol=2e-3)
         assert X[0].abs().max().item() == 0.0
 
-    @pytest.mark.parametrize("f", [16, 32, 64])
+    @pytest.mark.parametrize("f", [16, 32, 64, 128])
     @pytest.mark.parametrize("implicit", [False, True])
     def test_woodbury_seam(self, implicit, f):
         """Rows straddling WOODBURY_MAX_NNZ=32: small rows take the Woodbury
