@@ -149,10 +149,15 @@ def barrier() -> None:
 
 
 def max_scalar(x: float, device=None) -> float:
-    """MAX over ranks of a host scalar (for timing: take the slowest rank)."""
+    """MAX over ranks of a host scalar (for timing: take the slowest rank).
+
+    The staging tensor must live on the backend's device type: RCCL
+    process groups reject CPU tensors, so under nccl(=RCCL) the scalar
+    is staged through the current CUDA device."""
     if not is_distributed():
         return x
-    t = torch.tensor([x], dtype=torch.float64,
-                     device=device if device is not None else "cpu")
+    if device is None:
+        device = ("cuda" if dist.get_backend() == "nccl" else "cpu")
+    t = torch.tensor([x], dtype=torch.float64, device=device)
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     return float(t.item())
