@@ -22,7 +22,7 @@ extern "C" void launch_topk_score(
     const long long* ban_indptr, const int* ban_indices,
     float* out_val, int* out_idx,
     int B, long long N, int f, int K, int n_slices, int item_base,
-    hipStream_t stream);
+    unsigned long long* prof, hipStream_t stream);
 
 namespace {
 
@@ -109,7 +109,8 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
     torch::Tensor Xq, torch::Tensor Y, int64_t K, int64_t n_slices,
     c10::optional<torch::Tensor> item_mask,
     c10::optional<torch::Tensor> ban_indptr,
-    c10::optional<torch::Tensor> ban_indices, int64_t item_base) {
+    c10::optional<torch::Tensor> ban_indices, int64_t item_base,
+    c10::optional<torch::Tensor> prof) {
   check_cuda_f32(Xq, "Xq");
   check_cuda_f32(Y, "Y");
   const int64_t f = Y.size(1);
@@ -143,12 +144,21 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
   auto out_val = torch::empty({B, n_slices * 4 * K}, Xq.options());
   auto out_idx = torch::empty({B, n_slices * 4 * K},
                               Xq.options().dtype(torch::kInt32));
+  unsigned long long* prof_ptr = nullptr;
+  if (prof.has_value()) {
+    TORCH_CHECK(prof->is_cuda() && prof->is_contiguous() &&
+                    prof->scalar_type() == torch::kUInt64 &&
+                    prof->numel() >= 5,
+                "prof must be a contiguous u64[5] GPU tensor");
+    prof_ptr = reinterpret_cast<unsigned long long*>(
+        prof->data_ptr<uint64_t>());
+  }
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   launch_topk_score(Xq.data_ptr<float>(), Y.data_ptr<float>(), mask_ptr,
                     bi_ptr, bx_ptr, out_val.data_ptr<float>(),
                     out_idx.data_ptr<int>(), (int)B, (long long)N, (int)f,
-                    (int)K, (int)n_slices, (int)item_base, stream);
+                    (int)K, (int)n_slices, (int)item_base, prof_ptr, stream);
   C10_HIP_CHECK(hipGetLastError());
   return {out_val, out_idx};
 }
@@ -167,5 +177,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("topk_score", &topk_score, "Fused masked top-K scoring",
         py::arg("Xq"), py::arg("Y"), py::arg("K"), py::arg("n_slices") = 64,
         py::arg("item_mask") = py::none(), py::arg("ban_indptr") = py::none(),
-        py::arg("ban_indices") = py::none(), py::arg("item_base") = 0);
+        py::arg("ban_indices") = py::none(), py::arg("item_base") = 0,
+        py::arg("prof") = py::none());
 }

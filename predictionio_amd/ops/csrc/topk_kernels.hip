@@ -54,7 +54,11 @@ __device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
-template <int F>
+// PROF=true compiles in wall_clock64() phase accumulators (setup / item
+// staging / score+insert / writeback), summed per workgroup into
+// prof[5] = {setup, stage, score, write, blocks}. The live serving path
+// launches the PROF=false instantiation — zero probe cost.
+template <int F, bool PROF>
 __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     const float* __restrict__ Xq,        // B x F query vectors
     const float* __restrict__ Y,         // N x F item factors
@@ -63,7 +67,8 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     const int* __restrict__ ban_indices,         // sorted per user
     float* __restrict__ out_val,         // B x (n_slices*TK_WAVES) x K
     int* __restrict__ out_idx,
-    int B, long long N, int K, int n_slices, int item_base)
+    int B, long long N, int K, int n_slices, int item_base,
+    unsigned long long* prof)
 {
   constexpr int FP = F + 2;  // row stride in floats (float2-aligned, odd/2)
   // dynamic LDS: ys[TK_CHUNK][FP] | topv[TK_WAVES*TK_UPB][K] | topi[...]
@@ -75,6 +80,10 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+
+  const bool probe = PROF && tid == 0;
+  unsigned long long pt = 0, acc_setup = 0, acc_stage = 0, acc_score = 0;
+  if (probe) pt = wall_clock64();
 
   const int slice = blockIdx.x;
   const int ublock = blockIdx.y;
@@ -114,6 +123,12 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
   }
   __syncthreads();
 
+  if (probe) {
+    const unsigned long long now = wall_clock64();
+    acc_setup = now - pt;
+    pt = now;
+  }
+
   float th = -FLT_MAX;               // this lane's K-th best (this wave)
   float* tvu = topv + (wave * TK_UPB + lane) * K;
   int* tiu = topi + (wave * TK_UPB + lane) * K;
@@ -137,6 +152,11 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
       ys[c * FP + k] = Y[(base + c) * F + k];
     }
     __syncthreads();
+    if (probe) {
+      const unsigned long long now = wall_clock64();
+      acc_stage += now - pt;
+      pt = now;
+    }
 
     // wave w scores items [8w, 8w+8) of the chunk for all 64 users, in
     // GROUPS OF 4: serve PMC showed ~370 wait cycles per item — the
@@ -186,8 +206,14 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
       }
       }
     }
+    if (probe) {
+      const unsigned long long now = wall_clock64();
+      acc_score += now - pt;
+      pt = now;
+    }
   }
   __syncthreads();
+  if (probe) pt = wall_clock64();
 
   // ---- write out: group g = slice*TK_WAVES + wave, K entries per user
   for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
@@ -202,6 +228,13 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
       out_idx[o] = topi[(w * TK_UPB + u) * K + q];
     }
   }
+  if (probe) {
+    atomicAdd(&prof[0], acc_setup);
+    atomicAdd(&prof[1], acc_stage);
+    atomicAdd(&prof[2], acc_score);
+    atomicAdd(&prof[3], wall_clock64() - pt);
+    atomicAdd(&prof[4], 1ull);
+  }
 }
 
 extern "C" void launch_topk_score(
@@ -209,7 +242,7 @@ extern "C" void launch_topk_score(
     const long long* ban_indptr, const int* ban_indices,
     float* out_val, int* out_idx,
     int B, long long N, int f, int K, int n_slices, int item_base,
-    hipStream_t stream)
+    unsigned long long* prof, hipStream_t stream)
 {
   dim3 grid(n_slices, (B + TK_UPB - 1) / TK_UPB);
   dim3 block(256);
@@ -221,13 +254,23 @@ extern "C" void launch_topk_score(
     static bool attr_set_##FF = false;                                      \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
       hipFuncSetAttribute(                                                  \
-          reinterpret_cast<const void*>(&topk_score_kernel<FF>),            \
+          reinterpret_cast<const void*>(&topk_score_kernel<FF, false>),     \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);          \
+      hipFuncSetAttribute(                                                  \
+          reinterpret_cast<const void*>(&topk_score_kernel<FF, true>),      \
           hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);          \
       attr_set_##FF = true;                                                 \
     }                                                                       \
-    hipLaunchKernelGGL((topk_score_kernel<FF>), grid, block, lds_bytes,     \
-                       stream, Xq, Y, item_mask, ban_indptr, ban_indices,   \
-                       out_val, out_idx, B, N, K, n_slices, item_base);     \
+    if (prof != nullptr)                                                    \
+      hipLaunchKernelGGL((topk_score_kernel<FF, true>), grid, block,        \
+                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,   \
+                         ban_indices, out_val, out_idx, B, N, K, n_slices,  \
+                         item_base, prof);                                  \
+    else                                                                    \
+      hipLaunchKernelGGL((topk_score_kernel<FF, false>), grid, block,       \
+                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,   \
+                         ban_indices, out_val, out_idx, B, N, K, n_slices,  \
+                         item_base, nullptr);                               \
   } while (0)
   switch (f) {
     case 16: LAUNCH(16); break;
