@@ -42,6 +42,7 @@
 #define TK_UPB 64     // users per block (= lane count)
 
 typedef __attribute__((ext_vector_type(2))) float f32x2;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ bool in_sorted(const int* arr, int n, int x) {
   int lo = 0, hi = n - 1;
@@ -142,14 +143,77 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     ban = ban_indices + b0;
   }
 
+  // Item staging is software-pipelined through registers: while chunk i
+  // is scored out of LDS, chunk i+1's global loads are already in flight
+  // into stg[] (the phase probe measured un-overlapped staging at 33.6%
+  // of WG time — profiles/serve_phase_probe_r1.txt). Costs SREG = F/8
+  // VGPRs and no extra LDS. F=128's xreg pressure (128 VGPRs) cannot
+  // afford the pipeline registers, so it keeps the direct stage.
+  // float4 loads (Y rows are 16B-aligned) keep the in-flight register
+  // count low: a scalar-element pipeline spilled 68 B at F=64 — the 8
+  // loads' address registers outlived the fully unrolled score loop.
+  constexpr int Q4 = (TK_CHUNK * F) / 4;              // float4s per chunk
+  constexpr int SREG4 = Q4 >= 256 ? Q4 / 256 : 1;
+  constexpr bool PIPE = (F <= 64);
+  f32x4 stg4[PIPE ? SREG4 : 1];
+  if (PIPE) {
+    const int cn0 = (int)min((long long)TK_CHUNK, it1 - it0);
+    // uniform chunk base folded into the pointer: the per-thread offset
+    // stays 32-bit (c*F+k < 8K), so the loads take the scalar-base +
+    // v-offset form instead of spilling a 64-bit vector address
+    const float* Yb = Y + it0 * (long long)F;
+#pragma unroll
+    for (int r = 0; r < (PIPE ? SREG4 : 1); ++r) {
+      const int e4 = tid + r * 256;
+      const int c = e4 / (F / 4);
+      const int k = (e4 % (F / 4)) * 4;
+      stg4[r] = (e4 < Q4 && c < cn0)
+          ? *reinterpret_cast<const f32x4*>(&Yb[c * F + k])
+          : f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+  }
+
   for (long long base = it0; base < it1; base += TK_CHUNK) {
     const int cn = (int)min((long long)TK_CHUNK, it1 - base);
     __syncthreads();
-    // coalesced stage of cn item rows
-    for (int e = tid; e < cn * F; e += 256) {
-      const int c = e / F;
-      const int k = e % F;
-      ys[c * FP + k] = Y[(base + c) * F + k];
+    if (PIPE) {
+      // drain the in-flight registers into LDS (component stores: the
+      // FP=F+2 row stride is 8B- but not 16B-aligned), then issue the
+      // next chunk's loads so they fly during the score pass
+#pragma unroll
+      for (int r = 0; r < (PIPE ? SREG4 : 1); ++r) {
+        const int e4 = tid + r * 256;
+        if (e4 < Q4) {
+          const int c = e4 / (F / 4);
+          const int k = (e4 % (F / 4)) * 4;
+          float* dst = ys + c * FP + k;
+          dst[0] = stg4[r].x;
+          dst[1] = stg4[r].y;
+          dst[2] = stg4[r].z;
+          dst[3] = stg4[r].w;
+        }
+      }
+      const long long nbase = base + TK_CHUNK;
+      if (nbase < it1) {
+        const int cnn = (int)min((long long)TK_CHUNK, it1 - nbase);
+        const float* Yb = Y + nbase * (long long)F;
+#pragma unroll
+        for (int r = 0; r < (PIPE ? SREG4 : 1); ++r) {
+          const int e4 = tid + r * 256;
+          const int c = e4 / (F / 4);
+          const int k = (e4 % (F / 4)) * 4;
+          stg4[r] = (e4 < Q4 && c < cnn)
+              ? *reinterpret_cast<const f32x4*>(&Yb[c * F + k])
+              : f32x4{0.f, 0.f, 0.f, 0.f};
+        }
+      }
+    } else {
+      // coalesced stage of cn item rows
+      for (int e = tid; e < cn * F; e += 256) {
+        const int c = e / F;
+        const int k = e % F;
+        ys[c * FP + k] = Y[(base + c) * F + k];
+      }
     }
     __syncthreads();
     if (probe) {
