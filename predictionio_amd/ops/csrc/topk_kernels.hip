@@ -71,7 +71,8 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     int B, long long N, int K, int n_slices, int item_base,
     unsigned long long* prof)
 {
-  constexpr int FP = F + 2;  // row stride in floats (float2-aligned, odd/2)
+  constexpr int FP = F + 4;  // row stride: 16B-aligned so the score loop
+                             // reads ys rows as ds_read_b128 quads
   // dynamic LDS: ys[TK_CHUNK][FP] | topv[TK_WAVES*TK_UPB][K] | topi[...]
   extern __shared__ float lds[];
   float* ys = lds;
@@ -236,15 +237,23 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     for (int c0 = c_lo; c0 < c_hi; c0 += GRP) {
       const int gn = min(GRP, c_hi - c0);
       f32x2 a0 = {0.f, 0.f}, a1 = {0.f, 0.f};
-      const f32x2* yr0 = reinterpret_cast<const f32x2*>(ys + (c0 + 0) * FP);
-      const f32x2* yr1 = reinterpret_cast<const f32x2*>(
+      const f32x4* yr0 = reinterpret_cast<const f32x4*>(ys + (c0 + 0) * FP);
+      const f32x4* yr1 = reinterpret_cast<const f32x4*>(
           ys + (c0 + (GRP > 1 ? 1 : 0)) * FP);
       // FULLY unrolled: q must be a literal so xreg[] stays in registers
-      // (a capped unroll sent it to scratch).
+      // (a capped unroll sent it to scratch). ys rows are read in b128
+      // quads: halves the LDS instruction count in a loop that is
+      // ISSUE-bound (LDS reads and v_pk_fma share the SIMD issue slot).
 #pragma unroll
-      for (int q = 0; q < F / 2; ++q) {
-        a0 += xreg[q] * yr0[q];   // v_pk_fma_f32, indep item chains
-        if (GRP > 1) a1 += xreg[q] * yr1[q];
+      for (int q = 0; q < F / 4; ++q) {
+        const f32x4 y0 = yr0[q];
+        a0 += xreg[2 * q] * f32x2{y0.x, y0.y};
+        a0 += xreg[2 * q + 1] * f32x2{y0.z, y0.w};
+        if (GRP > 1) {
+          const f32x4 y1 = yr1[q];
+          a1 += xreg[2 * q] * f32x2{y1.x, y1.y};
+          a1 += xreg[2 * q + 1] * f32x2{y1.z, y1.w};
+        }
       }
       float sg[2] = {a0.x + a0.y, a1.x + a1.y};
 #pragma unroll
@@ -313,7 +322,7 @@ extern "C" void launch_topk_score(
 #define LAUNCH(FF)                                                          \
   do {                                                                      \
     const size_t lds_bytes =                                                \
-        sizeof(float) * (TK_CHUNK * (FF + 2)) +                             \
+        sizeof(float) * (TK_CHUNK * (FF + 4)) +                             \
         (sizeof(float) + sizeof(int)) * TK_WAVES * TK_UPB * K;              \
     static bool attr_set_##FF = false;                                      \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
