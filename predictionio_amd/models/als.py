@@ -128,7 +128,10 @@ class ALSTrainer:
             if self.device.type == "cuda":
                 torch.cuda.synchronize()
             t1 = time.time()
-        yty = als_ops.gramian(fixed_full) if self.p.implicit else None
+        # YtY from the LOCAL shard + an FxF all-reduce: the post-gather
+        # Gramian would redo a world-size-times-larger GEMM on every rank
+        yty = (pdist.all_reduce_sum(als_ops.gramian(fixed_local))
+               if self.p.implicit else None)
         indptr, indices, values = csr
         out = als_ops.als_solve(
             indptr, indices, values, fixed_full, YtY=yty,

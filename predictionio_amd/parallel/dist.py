@@ -143,6 +143,17 @@ def exchange_triples(rows: torch.Tensor, cols: torch.Tensor,
     return a2a(rows_s), a2a(cols_s), a2a(vals_s)
 
 
+def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
+    """In-place SUM all-reduce (no-op when not distributed). Used for
+    tiny reductions like the FxF Gramian: summing local X^T X across
+    ranks replaces an F x F GEMM over the full gathered matrix — the
+    redundant post-gather Gramian grows with world size, the local one
+    does not (NOTES.md "8-GPU scaling model" lever c)."""
+    if is_distributed():
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
 def barrier() -> None:
     if is_distributed():
         dist.barrier()
