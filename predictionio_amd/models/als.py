@@ -36,6 +36,9 @@ class ALSParams:
     alpha: float = 1.0          # implicit confidence weight
     implicit: bool = False
     seed: Optional[int] = None
+    gather_dtype: Optional[str] = None  # "bf16" halves multi-GPU factor
+                                        # gather bytes; numerics study in
+                                        # profiles/bf16_numerics_study.txt
 
 
 class ALSTrainer:
@@ -123,7 +126,9 @@ class ALSTrainer:
             if self.device.type == "cuda":
                 torch.cuda.synchronize()
             t0 = time.time()
-        fixed_full = pdist.all_gather_rows(fixed_local, n_fixed)
+        wire = (torch.bfloat16 if self.p.gather_dtype == "bf16" else None)
+        fixed_full = pdist.all_gather_rows(fixed_local, n_fixed,
+                                           wire_dtype=wire)
         if timing:
             if self.device.type == "cuda":
                 torch.cuda.synchronize()

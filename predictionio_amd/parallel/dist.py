@@ -58,14 +58,25 @@ def block_bounds(n: int, world: int, rank: int) -> Tuple[int, int]:
     return lo, hi
 
 
-def all_gather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
+def all_gather_rows(local: torch.Tensor, n_total: int,
+                    wire_dtype: torch.dtype | None = None) -> torch.Tensor:
     """All-gather row-sharded [n_local, f] tensors into [n_total, f].
 
     Shards may be uneven (block_bounds); pads to the max shard for
     all_gather_into_tensor (single fused RCCL call over xGMI), then
-    reassembles. No-op when not distributed."""
+    reassembles. No-op when not distributed.
+
+    wire_dtype: optional reduced dtype for the collective only (e.g.
+    torch.bfloat16 halves the 25.6 GB 8-GPU item-factor gather); the
+    result is cast back to local.dtype. Numerics: bf16 rounding of
+    factor VALUES at this boundary tracks fp32 ALS to ~1e-5 relative
+    objective (profiles/bf16_numerics_study.txt). Off by default —
+    opt in via ALSParams.gather_dtype."""
     if not is_distributed():
         return local
+    out_dtype = local.dtype
+    if wire_dtype is not None and wire_dtype != local.dtype:
+        local = local.to(wire_dtype)
     world = get_world_size()
     f = local.shape[1]
     max_rows = (n_total + world - 1) // world
@@ -85,7 +96,8 @@ def all_gather_rows(local: torch.Tensor, n_total: int) -> torch.Tensor:
     for r in range(world):
         lo, hi = block_bounds(n_total, world, r)
         pieces.append(out[r * max_rows: r * max_rows + (hi - lo)])
-    return torch.cat(pieces, dim=0) if world > 1 else out[:n_total]
+    res = torch.cat(pieces, dim=0) if world > 1 else out[:n_total]
+    return res.to(out_dtype) if res.dtype != out_dtype else res
 
 
 def exchange_triples(rows: torch.Tensor, cols: torch.Tensor,

@@ -54,6 +54,22 @@ def _gather_uneven(rank, world):
     return full.numpy().tolist()
 
 
+def _gather_bf16_wire(rank, world):
+    """wire_dtype=bf16 gather: result equals the bf16-rounded shards."""
+    from predictionio_amd.parallel import dist as pdist
+    n_total = 7
+    lo, hi = pdist.block_bounds(n_total, world, rank)
+    g = torch.Generator().manual_seed(5 + rank)
+    local = torch.randn((hi - lo, 8), generator=g)
+    full = pdist.all_gather_rows(local, n_total,
+                                 wire_dtype=torch.bfloat16)
+    assert full.dtype == torch.float32 and full.shape == (n_total, 8)
+    # this rank's own block must round-trip exactly through bf16
+    expect = local.to(torch.bfloat16).float()
+    assert torch.equal(full[lo:hi], expect)
+    return True
+
+
 def _dist_als(rank, world):
     """Distributed ALS on 2 ranks must equal single-process ALS."""
     from predictionio_amd.models.als import ALSParams, ALSTrainer
@@ -173,6 +189,10 @@ class TestExchangeTriples:
 
 
 class TestBenchDistPath:
+    def test_gather_bf16_wire(self):
+        res = _spawn("_gather_bf16_wire", port=29621)
+        assert all(res.values())
+
     def test_bench_shard_setup(self):
         res = _spawn("_bench_shard_path", port=29619)
         assert res[0] == [16, 12] and res[1] == [16, 12]
