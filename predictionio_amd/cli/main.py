@@ -96,10 +96,14 @@ def build(engine_dir):
 def train(engine_dir, variant, batch, skip_sanity_check, gpus):
     """Train an engine instance (RunWorkflow → CreateWorkflow.main)."""
     if gpus > 1:
+        import socket
         import subprocess
+        with socket.socket() as _s:  # free rendezvous port (29500 default
+            _s.bind(("127.0.0.1", 0))  # collides with other torchruns)
+            port = _s.getsockname()[1]
         cmd = [sys.executable, "-m", "torch.distributed.run",
                "--nnodes=1", f"--nproc-per-node={gpus}",
-               "--master-addr", "127.0.0.1",
+               "--master-addr", "127.0.0.1", "--master-port", str(port),
                "-m", "predictionio_amd.cli.train_main",
                "--engine-dir", engine_dir, "--variant", variant,
                "--batch", batch]

@@ -137,6 +137,60 @@ class TestQuickstart:
                              "attr2": 9}).json() == {"label": 1.0}
 
 
+class TestMultiProcessTrain:
+    def test_pio_train_gpus2_cpu(self, mem_storage, tmp_path):
+        """`pio train --gpus 2` spawns torchrun (gloo on CPU): both ranks
+        read the shared sqlite store, shard the ALS solve, rank 0
+        persists — then the model deploys and answers queries."""
+        import json as js
+        from click.testing import CliRunner
+        from predictionio_amd.cli.main import cli
+        runner = CliRunner()
+        r = runner.invoke(cli, ["app", "new", "DistApp",
+                                "--access-key", "DKEY"])
+        assert r.exit_code == 0
+        from predictionio_amd.data import storage
+        app_id = storage.get_meta_data_apps().get_by_name("DistApp").id
+        le = storage.get_l_events()
+        import random
+        from datetime import datetime, timezone
+        from predictionio_amd.data.events import DataMap, Event
+        rng = random.Random(3)
+        for u in range(20):
+            liked = [i for i in range(12) if i % 2 == u % 2]
+            for i in rng.sample(liked, 4):
+                le.insert(Event(
+                    event="rate", entity_type="user", entity_id=f"u{u}",
+                    target_entity_type="item", target_entity_id=f"i{i}",
+                    properties=DataMap({"rating": rng.uniform(3.5, 5.0)}),
+                    event_time=datetime.now(timezone.utc)), app_id)
+        engine_dir = tmp_path / "eng"
+        r = runner.invoke(cli, ["template", "get", "recommendation",
+                                str(engine_dir)])
+        assert r.exit_code == 0
+        variant = js.loads((engine_dir / "engine.json").read_text())
+        variant["datasource"]["params"]["appName"] = "DistApp"
+        variant["algorithms"][0]["params"].update(
+            {"rank": 8, "numIterations": 3})
+        (engine_dir / "engine.json").write_text(js.dumps(variant))
+        r = runner.invoke(cli, ["train", "--engine-dir", str(engine_dir),
+                                "--gpus", "2"])
+        assert r.exit_code == 0, r.output
+        # the COMPLETED instance was written by rank 0 of the torchrun job
+        insts = storage.get_meta_data_engine_instances()
+        done = [i for i in insts.get_all() if i.status == "COMPLETED"]
+        assert done, "no COMPLETED engine instance from distributed train"
+        from predictionio_amd.server.queryserver import (
+            ServerConfig, create_app as qs_app,
+        )
+        qs = TestClient(qs_app(ServerConfig(
+            engine_factory=variant["engineFactory"],
+            engine_variant=variant["id"])))
+        resp = qs.post("/queries.json", json={"user": "u1", "num": 3})
+        assert resp.status_code == 200
+        assert len(resp.json()["itemScores"]) == 3
+
+
 class TestFsspecModels:
     def test_model_blob_roundtrip(self, tmp_path, monkeypatch):
         from predictionio_amd.data import storage
