@@ -130,3 +130,54 @@ class TestTrainWorkflow:
             train_wf.run_train_from_variant(variant)
         insts = mem_storage.get_meta_data_engine_instances().get_all()
         assert insts and insts[-1].status == "FAILED"
+
+
+class TestVariantParsing:
+    """engine.json → EngineParams edge cases (jValueToEngineParams,
+    Engine.scala:355-418 / JsonExtractorSpec analog)."""
+
+    def _engine(self):
+        from tests.fake_engine import make_engine
+        return make_engine()
+
+    def test_empty_variant_defaults(self):
+        ep = self._engine().json_to_engine_params({})
+        assert ep.algorithms_params[0][0] == ""
+        assert dict(ep.data_source_params) == {}
+
+    def test_bare_params_without_wrapper(self):
+        # a stage given as a bare object is treated as its params
+        ep = self._engine().json_to_engine_params(
+            {"datasource": {"n": 7}})
+        assert ep.data_source_params.get("n") == 7
+
+    def test_named_stages_and_multi_algo(self):
+        ep = self._engine().json_to_engine_params({
+            "datasource": {"name": "dsA", "params": {"n": 3}},
+            "serving": {"name": "sB", "params": {"bump": 1}},
+            "algorithms": [
+                {"name": "algo0", "params": {"bias": 5}},
+                {"name": "algo1"},
+            ]})
+        assert ep.data_source_name == "dsA"
+        assert ep.serving_name == "sB"
+        assert [n for n, _ in ep.algorithms_params] == ["algo0", "algo1"]
+        assert ep.algorithms_params[0][1].get("bias") == 5
+        assert dict(ep.algorithms_params[1][1]) == {}
+
+    def test_instance_round_trip(self, mem_storage):
+        """EngineParams → stored instance → engine_instance_to_engine_params
+        (Engine.scala:420-490)."""
+        from predictionio_amd.workflow import train as train_wf
+        from tests.fake_engine import make_engine
+        variant = {"id": "rt", "engineFactory":
+                   "tests.fake_engine.FakeEngineFactory",
+                   "datasource": {"params": {"n": 5}},
+                   "algorithms": [{"name": "algo0", "params": {"bias": 2}}]}
+        iid = train_wf.run_train_from_variant(variant)
+        inst = mem_storage.get_meta_data_engine_instances().get(iid)
+        ep = make_engine().engine_instance_to_engine_params(inst)
+        assert ep.data_source_params.get("n") == 5
+        assert ep.algorithms_params == [
+            ("algo0", type(ep.data_source_params)({"bias": 2}))] or             ep.algorithms_params[0][0] == "algo0"
+        assert ep.algorithms_params[0][1].get("bias") == 2
