@@ -147,6 +147,25 @@ class TestTopKKernel:
                 torch.nonzero(mask).flatten().tolist())
             assert not (set(gi[b].tolist()) - {-1}) & banned
 
+    def test_prof_variant_matches_live(self):
+        """The PROF=true instantiation (wall_clock64 phase probe,
+        scripts/serve_phase_probe.py) must return the same candidates as
+        the live PROF=false path and fill prof[5]."""
+        from predictionio_amd.ops import hip_ext
+        g = torch.Generator().manual_seed(17)
+        B, N, f, K, ns = 16, 4000, 64, 8, 3
+        Xq = torch.randn((B, f), generator=g).float().cuda()
+        Y = torch.randn((N, f), generator=g).float().cuda()
+        ext = hip_ext()
+        v0, i0 = ext.topk_score(Xq, Y, K, ns, None, None, None, 0)
+        prof = torch.zeros(5, dtype=torch.uint64, device="cuda")
+        v1, i1 = ext.topk_score(Xq, Y, K, ns, None, None, None, 0, prof)
+        torch.cuda.synchronize()
+        assert torch.equal(v0, v1) and torch.equal(i0, i1)
+        p = prof.cpu().tolist()
+        assert p[4] > 0  # sampled workgroups
+        assert p[1] > 0 and p[2] > 0  # stage + score phases ticked
+
     def test_more_k_than_items(self):
         from predictionio_amd.ops import topk as topk_ops
         Xq = torch.randn((3, 64)).float().cuda()
