@@ -77,11 +77,15 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
   extern __shared__ float lds[];
   float* ys = lds;
   float* topv = ys + TK_CHUNK * FP;
-  int* topi = reinterpret_cast<int*>(topv + TK_WAVES * TK_UPB * K);
+  int* topi = reinterpret_cast<int*>(topv + TK_WAVES * TK_UPB * (K + 1));
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  // K-list row stride padded by one: lane-consecutive K=20 lists had an
+  // 8-way LDS bank conflict (gcd(20,32)=4; PMC: 485k conflicts/wave =
+  // 34% of LDS instructions); stride 21 is coprime with the 32 banks.
+  const int KP = K + 1;
 
   const bool probe = PROF && tid == 0;
   unsigned long long pt = 0, acc_setup = 0, acc_stage = 0, acc_score = 0;
@@ -119,7 +123,7 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
   __syncthreads();
 
   // ---- init this block's top-K lists
-  for (int e = tid; e < TK_WAVES * TK_UPB * K; e += 256) {
+  for (int e = tid; e < TK_WAVES * TK_UPB * KP; e += 256) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
@@ -132,8 +136,8 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
   }
 
   float th = -FLT_MAX;               // this lane's K-th best (this wave)
-  float* tvu = topv + (wave * TK_UPB + lane) * K;
-  int* tiu = topi + (wave * TK_UPB + lane) * K;
+  float* tvu = topv + (wave * TK_UPB + lane) * KP;
+  int* tiu = topi + (wave * TK_UPB + lane) * KP;
 
   // this lane's banned list
   const int* ban = nullptr;
@@ -297,8 +301,8 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
     if (gu < B) {
       const long long g = (long long)slice * TK_WAVES + w;
       const long long o = (gu * n_slices * TK_WAVES + g) * K + q;
-      out_val[o] = topv[(w * TK_UPB + u) * K + q];
-      out_idx[o] = topi[(w * TK_UPB + u) * K + q];
+      out_val[o] = topv[(w * TK_UPB + u) * KP + q];
+      out_idx[o] = topi[(w * TK_UPB + u) * KP + q];
     }
   }
   if (probe) {
@@ -321,9 +325,12 @@ extern "C" void launch_topk_score(
   dim3 block(256);
 #define LAUNCH(FF)                                                          \
   do {                                                                      \
-    const size_t lds_bytes =                                                \
+    size_t lds_bytes =                                                      \
         sizeof(float) * (TK_CHUNK * (FF + 4)) +                             \
-        (sizeof(float) + sizeof(int)) * TK_WAVES * TK_UPB * K;              \
+        (sizeof(float) + sizeof(int)) * TK_WAVES * TK_UPB * (K+1);        \
+    /* small-K floor: the query-staging pass writes TK_UPB rows     */  \
+    const size_t stage_need = sizeof(float) * TK_UPB * (FF + 4);        \
+    if (lds_bytes < stage_need) lds_bytes = stage_need;                 \
     static bool attr_set_##FF = false;                                      \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                          \
       hipFuncSetAttribute(                                                  \

@@ -147,6 +147,24 @@ class TestTopKKernel:
                 torch.nonzero(mask).flatten().tolist())
             assert not (set(gi[b].tolist()) - {-1}) & banned
 
+    def test_small_k_full_block(self):
+        """K small enough that ys+lists < the 17.4 KB query-staging pass,
+        with B=64 filling the user block: high lanes' staged queries used
+        to land beyond the dynamic LDS allocation (OOB ds_writes are
+        dropped, the transpose read zeros) — the launcher now floors the
+        allocation at the staging footprint."""
+        from predictionio_amd.ops import topk as topk_ops
+        g = torch.Generator().manual_seed(23)
+        B, N, f = 64, 3000, 64
+        Xq = torch.randn((B, f), generator=g).float()
+        Y = torch.randn((N, f), generator=g).float()
+        for K in (1, 4, 8):
+            rv, ri = topk_ops.topk_score_ref(Xq, Y, K)
+            gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K,
+                                         n_slices=5)
+            gv, gi = gv.cpu(), gi.cpu()
+            assert torch.allclose(gv, rv, atol=1e-4, rtol=1e-4),                 f"K={K}: max diff {(gv - rv).abs().max()} "                 f"(worst user {int((gv - rv).abs().max(1)[0].argmax())})"
+
     def test_prof_variant_matches_live(self):
         """The PROF=true instantiation (wall_clock64 phase probe,
         scripts/serve_phase_probe.py) must return the same candidates as
