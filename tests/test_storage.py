@@ -224,3 +224,46 @@ class TestModelsBackendMatrix:
         assert m.delete("x1") is True
         assert m.delete("x1") is False
         storage.reset()
+
+
+class TestConcurrentAccess:
+    """The event server runs handlers on a thread pool (uvicorn): the
+    sqlite DAO uses a connection per thread (file-backed) / one locked
+    connection (:memory:) — hammer both from many threads."""
+
+    def _hammer(self, le, app_id, threads=8, per=25):
+        import threading
+        from datetime import datetime, timezone
+        from predictionio_amd.data.events import DataMap, Event
+        errors = []
+
+        def work(t):
+            try:
+                for k in range(per):
+                    eid = le.insert(Event(
+                        event="rate", entity_type="user",
+                        entity_id=f"u{t}", target_entity_type="item",
+                        target_entity_id=f"i{k}",
+                        properties=DataMap({"rating": k % 5 + 1}),
+                        event_time=datetime.now(timezone.utc)), app_id)
+                    assert le.get(eid, app_id) is not None
+                    le.find(app_id, entity_id=f"u{t}", limit=5)
+            except Exception as e:  # noqa: BLE001
+                errors.append(repr(e))
+
+        ts = [threading.Thread(target=work, args=(t,))
+              for t in range(threads)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=60)
+        assert not errors, errors[:3]
+        evs = list(le.find(app_id, limit=-1))
+        assert len(evs) == threads * per
+
+    def test_sqlite_file_threaded(self, mem_storage):
+        from predictionio_amd.data.storage.base import App
+        app_id = mem_storage.get_meta_data_apps().insert(App(0, "ThrApp"))
+        le = mem_storage.get_l_events()
+        le.init(app_id)
+        self._hammer(le, app_id)
