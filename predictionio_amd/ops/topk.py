@@ -66,10 +66,15 @@ def topk_score_ref(Xq, Y, K, item_mask=None, ban_indptr=None,
             banned = ban_indices[ip[b]:ip[b + 1]].long()
             if banned.numel():
                 scores[b, banned] = float("-inf")
-    K = min(K, scores.shape[1])
-    vals, idxs = torch.topk(scores, K, dim=1)
+    Kc = min(K, scores.shape[1])
+    vals, idxs = torch.topk(scores, Kc, dim=1)
     idxs = idxs.clone()
     idxs[vals == float("-inf")] = -1
+    if Kc < K:  # pad to K like the GPU kernel (empty slots = -inf / -1)
+        B = scores.shape[0]
+        vals = torch.cat([vals, torch.full((B, K - Kc), float("-inf"))], 1)
+        idxs = torch.cat([idxs, torch.full((B, K - Kc), -1,
+                                           dtype=idxs.dtype)], 1)
     return vals, idxs
 
 
