@@ -86,11 +86,14 @@ class TestCSRProperties:
         rebuilt = defaultdict(list)
         for r in range(n_rows):
             for j in range(int(indptr[r]), int(indptr[r + 1])):
-                rebuilt[r].append((int(indices[j]),
-                                   round(float(values[j]), 4)))
+                rebuilt[r].append((int(indices[j]), float(values[j])))
         expect = defaultdict(list)
         for r, c, v in triples:
-            expect[r].append((c, round(v, 4)))
+            # build_csr only permutes: values must round-trip BIT-exactly
+            # through their fp32 representation (round(x, 4) straddled
+            # fp32 rounding boundaries and was flaky)
+            expect[r].append((c, float(torch.tensor(v,
+                                                    dtype=torch.float32))))
         for r in range(n_rows):
             assert sorted(rebuilt[r]) == sorted(expect[r])
 
@@ -121,9 +124,93 @@ class TestCSRProperties:
         cols = torch.tensor([t[1] for t in triples], dtype=torch.int32)
         vals = torch.tensor([t[2] for t in triples], dtype=torch.float32)
         r2, c2, v2 = aggregate_ratings(rows, cols, vals, 5, "latest")
-        got = {(int(r), int(c)): round(float(v), 4)
+        got = {(int(r), int(c)): float(v)
                for r, c, v in zip(r2, c2, v2)}
         expect = {}
-        for r, c, v in triples:  # later entries win
-            expect[(r, c)] = round(v, 4)
+        for r, c, v in triples:  # later entries win; fp32-exact passthrough
+            expect[(r, c)] = float(torch.tensor(v, dtype=torch.float32))
         assert got == expect
+
+
+class TestFindFilterProperties:
+    """le.find's 9-dimension filter vs a brute-force list filter on
+    arbitrary event sets and filter combos (LEvents.scala:188-200
+    semantics: UNSET=any, None=absent, str=equal)."""
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.data())
+    def test_matches_bruteforce(self, data):
+        import itertools
+        from predictionio_amd.data import storage as storage_mod
+        from predictionio_amd.data.storage.base import UNSET
+        from predictionio_amd.data.storage.sqlite import (
+            SQLiteClient, SQLiteLEvents,
+        )
+        le = SQLiteLEvents(SQLiteClient(":memory:"))
+        le.init(1)
+        n = data.draw(st.integers(0, 30))
+        etypes = ["user", "item"]
+        enames = ["rate", "buy", "$set"]
+        evs = []
+        for k in range(n):
+            has_tgt = data.draw(st.booleans())
+            e = Event(
+                event=data.draw(st.sampled_from(enames)),
+                entity_type=data.draw(st.sampled_from(etypes)),
+                entity_id=data.draw(st.sampled_from(["a", "b", "c"])),
+                target_entity_type="item" if has_tgt else None,
+                target_entity_id=data.draw(
+                    st.sampled_from(["x", "y"])) if has_tgt else None,
+                properties=DataMap({}),
+                event_time=T0 + timedelta(seconds=data.draw(
+                    st.integers(0, 50))))
+            le.insert(e, 1)
+            evs.append(e)
+        kw = {}
+        if data.draw(st.booleans()):
+            kw["entity_type"] = data.draw(st.sampled_from(etypes))
+        if data.draw(st.booleans()):
+            kw["entity_id"] = data.draw(st.sampled_from(["a", "b", "z"]))
+        if data.draw(st.booleans()):
+            kw["event_names"] = data.draw(
+                st.lists(st.sampled_from(enames), min_size=1,
+                         max_size=2, unique=True))
+        if data.draw(st.booleans()):
+            kw["start_time"] = T0 + timedelta(
+                seconds=data.draw(st.integers(0, 50)))
+        if data.draw(st.booleans()):
+            kw["until_time"] = T0 + timedelta(
+                seconds=data.draw(st.integers(0, 50)))
+        tt = data.draw(st.sampled_from(["unset", "none", "item"]))
+        if tt != "unset":
+            kw["target_entity_type"] = None if tt == "none" else tt
+        rev = data.draw(st.booleans())
+        got = list(le.find(1, reversed=rev, **kw))
+
+        def keep(e):
+            if "entity_type" in kw and e.entity_type != kw["entity_type"]:
+                return False
+            if "entity_id" in kw and e.entity_id != kw["entity_id"]:
+                return False
+            if "event_names" in kw and e.event not in kw["event_names"]:
+                return False
+            if "start_time" in kw and e.event_time < kw["start_time"]:
+                return False
+            if "until_time" in kw and e.event_time >= kw["until_time"]:
+                return False
+            if tt == "none" and e.target_entity_type is not None:
+                return False
+            if tt == "item" and e.target_entity_type != "item":
+                return False
+            return True
+
+        expect = [e for e in evs if keep(e)]
+        assert len(got) == len(expect)
+        # ordering: eventTime ascending (descending when reversed)
+        times = [e.event_time for e in got]
+        assert times == sorted(times, reverse=rev)
+        got_keys = sorted((e.event, e.entity_id,
+                           e.event_time.isoformat()) for e in got)
+        exp_keys = sorted((e.event, e.entity_id,
+                           e.event_time.isoformat()) for e in expect)
+        assert got_keys == exp_keys
