@@ -190,7 +190,7 @@ class TestExchangeTriples:
 
 class TestBenchDistPath:
     def test_gather_bf16_wire(self):
-        res = _spawn("_gather_bf16_wire", port=29621)
+        res = _spawn("_gather_bf16_wire", port=29627)
         assert all(res.values())
 
     def test_bench_shard_setup(self):
