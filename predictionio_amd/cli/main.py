@@ -141,8 +141,12 @@ def eval_cmd(evaluation, generator, engine_dir):
 @click.option("--event-server-ip", default="localhost")
 @click.option("--event-server-port", default=7070)
 @click.option("--accesskey", default=None)
+@click.option("--batch-window-ms", default=0.0,
+              help="coalesce concurrent queries for this window into one "
+                   "fused batch_predict launch (0 = per-request)")
+@click.option("--max-batch", default=64)
 def deploy(engine_dir, variant, ip, port, feedback, event_server_ip,
-           event_server_port, accesskey):
+           event_server_port, accesskey, batch_window_ms, max_batch):
     """Deploy the latest completed instance behind /queries.json
     (commands/Engine.deploy :208-245 → CreateServer)."""
     _add_engine_dir(engine_dir)
@@ -153,7 +157,8 @@ def deploy(engine_dir, variant, ip, port, feedback, event_server_ip,
         engine_variant=v.get("id", "default"),
         ip=ip, port=port, feedback=feedback,
         event_server_uri=f"http://{event_server_ip}:{event_server_port}",
-        access_key=accesskey)
+        access_key=accesskey, batch_window_ms=batch_window_ms,
+        max_batch=max_batch)
     run(cfg)
 
 

@@ -66,6 +66,14 @@ class ServerConfig:
     app_name: Optional[str] = None
     engine_instance_id: Optional[str] = None  # None = latest completed
     log_url: Optional[str] = None  # remote error log POST target
+    # Dynamic micro-batching (SURVEY.md §2.8 serving concurrency): when
+    # batch_window_ms > 0, concurrent /queries.json requests are
+    # coalesced for up to that window (or max_batch) and dispatched
+    # through Algorithm.batch_predict — for the GPU templates that is
+    # ONE fused top-K kernel launch for the whole micro-batch instead of
+    # a launch per request. 0 = per-request path (reference semantics).
+    batch_window_ms: float = 0.0
+    max_batch: int = 64
 
 
 @dataclass
@@ -115,6 +123,70 @@ def _load_state(config: ServerConfig):
         serving=engine._serving(ep), algorithms=engine._algorithms(ep))
 
 
+class _MicroBatcher:
+    """Coalesces concurrent queries into Algorithm.batch_predict calls.
+
+    The queue lives on the event loop; the batch itself (supplement →
+    batch_predict per algorithm → serve per query) runs on a worker
+    thread so model compute never blocks the loop. Failures fan the
+    exception out to every waiting request in the batch."""
+
+    def __init__(self, holder, max_batch: int, window_s: float):
+        self.holder = holder
+        self.max_batch = max_batch
+        self.window = window_s
+        self.queue = None
+        self._task = None
+
+    def start(self):
+        import asyncio
+        self.queue = asyncio.Queue()
+        self._task = asyncio.get_running_loop().create_task(self._drain())
+
+    async def submit(self, query):
+        import asyncio
+        fut = asyncio.get_running_loop().create_future()
+        await self.queue.put((query, fut))
+        return await fut
+
+    async def _drain(self):
+        import asyncio
+        loop = asyncio.get_running_loop()
+        while True:
+            batch = [await self.queue.get()]
+            deadline = loop.time() + self.window
+            while len(batch) < self.max_batch:
+                left = deadline - loop.time()
+                if left <= 0:
+                    break
+                try:
+                    batch.append(await asyncio.wait_for(
+                        self.queue.get(), left))
+                except asyncio.TimeoutError:
+                    break
+            await loop.run_in_executor(None, self._run_batch, loop, batch)
+
+    def _run_batch(self, loop, batch):
+        s = self.holder["st"]
+        try:
+            supplemented = [s.serving.supplement(q) for q, _ in batch]
+            indexed = list(enumerate(supplemented))
+            per_algo = []
+            for a, m in zip(s.algorithms, s.models):
+                out = dict(a.batch_predict(m, indexed))
+                per_algo.append([out[i] for i in range(len(batch))])
+            for i, (q, fut) in enumerate(batch):
+                res = s.serving.serve(q, [pa[i] for pa in per_algo])
+                loop.call_soon_threadsafe(
+                    lambda f=fut, r=res: (not f.done()) and f.set_result(r))
+        except Exception as e:  # noqa: BLE001
+            logger.exception("micro-batch failed")
+            for _, fut in batch:
+                loop.call_soon_threadsafe(
+                    lambda f=fut, err=e: (not f.done())
+                    and f.set_exception(RuntimeError(str(err))))
+
+
 def create_app(config: ServerConfig,
                plugins: Optional[List[EngineServerPlugin]] = None,
                state: Optional[_ServingState] = None) -> FastAPI:
@@ -127,6 +199,14 @@ def create_app(config: ServerConfig,
     st = state if state is not None else _load_state(config)
     holder = {"st": st}
     lock = threading.Lock()
+    batcher = None
+    if config.batch_window_ms > 0:
+        batcher = _MicroBatcher(holder, config.max_batch,
+                                config.batch_window_ms / 1000.0)
+
+        @app.on_event("startup")
+        async def _start_batcher():
+            batcher.start()
 
     def _post_feedback(query_json: dict, prediction_json: dict,
                        pr_id: str) -> None:
@@ -207,10 +287,13 @@ def create_app(config: ServerConfig,
             query = s.algorithms[0].query_from_json(query_json) \
                 if hasattr(s.algorithms[0], "query_from_json") \
                 else query_json
-            supplemented = s.serving.supplement(query)
-            predictions = [a.predict(m, supplemented)
-                           for a, m in zip(s.algorithms, s.models)]
-            prediction = s.serving.serve(query, predictions)
+            if batcher is not None:
+                prediction = await batcher.submit(query)
+            else:
+                supplemented = s.serving.supplement(query)
+                predictions = [a.predict(m, supplemented)
+                               for a, m in zip(s.algorithms, s.models)]
+                prediction = s.serving.serve(query, predictions)
         except Exception as e:
             logger.exception("query failed")
             if config.log_url:

@@ -146,3 +146,62 @@ class TestStopAuth:
         monkeypatch.setattr(os, "kill", lambda *a: None)
         assert c.post("/stop?accessKey=SECRET").status_code == 200
         time.sleep(0.4)
+
+
+class TestMicroBatching:
+    """Dynamic micro-batching (ServerConfig.batch_window_ms): queries are
+    routed through Algorithm.batch_predict — one fused launch per
+    coalesced batch — with per-request results identical to the
+    per-request path."""
+
+    def _app(self, mem_storage, window_ms=5.0):
+        _train("vb")
+        from predictionio_amd.server.queryserver import (
+            ServerConfig, create_app,
+        )
+        cfg = ServerConfig(engine_factory=FACTORY, engine_variant="vb",
+                           batch_window_ms=window_ms, max_batch=8)
+        return create_app(cfg)
+
+    def test_single_query_via_batch_path(self, mem_storage, monkeypatch):
+        from tests.fake_engine import JsonAlgo
+        calls = {"batch": 0, "single": 0}
+        orig = JsonAlgo.batch_predict
+
+        def counting_batch(self, model, queries):
+            calls["batch"] += 1
+            return orig(self, model, queries)
+
+        monkeypatch.setattr(JsonAlgo, "batch_predict", counting_batch)
+        monkeypatch.setattr(
+            JsonAlgo, "predict",
+            lambda self, m, q: calls.__setitem__("single", 1) or
+            {"result": m + q["x"]})
+        app = self._app(mem_storage)
+        with TestClient(app) as c:  # context manager runs startup events
+            r = c.post("/queries.json", json={"x": 5})
+            assert r.status_code == 200
+            assert r.json() == {"result": 6 + 5}
+        assert calls["batch"] >= 1
+
+    def test_concurrent_queries_all_correct(self, mem_storage):
+        import threading
+        app = self._app(mem_storage, window_ms=10.0)
+        results = {}
+        with TestClient(app) as c:
+            def one(x):
+                results[x] = c.post("/queries.json",
+                                    json={"x": x}).json()
+            ts = [threading.Thread(target=one, args=(x,))
+                  for x in range(12)]
+            for t in ts:
+                t.start()
+            for t in ts:
+                t.join(timeout=30)
+        assert results == {x: {"result": 6 + x} for x in range(12)}
+
+    def test_batch_error_propagates(self, mem_storage):
+        app = self._app(mem_storage)
+        with TestClient(app) as c:
+            r = c.post("/queries.json", json={"wrong_key": 1})
+            assert r.status_code == 500
