@@ -204,10 +204,6 @@ def create_app(config: ServerConfig,
         batcher = _MicroBatcher(holder, config.max_batch,
                                 config.batch_window_ms / 1000.0)
 
-        @app.on_event("startup")
-        async def _start_batcher():
-            batcher.start()
-
     def _post_feedback(query_json: dict, prediction_json: dict,
                        pr_id: str) -> None:
         """Async predict-event feedback (CreateServer.scala:527-589)."""
@@ -288,6 +284,8 @@ def create_app(config: ServerConfig,
                 if hasattr(s.algorithms[0], "query_from_json") \
                 else query_json
             if batcher is not None:
+                if batcher.queue is None:  # lazily bind to the loop
+                    batcher.start()
                 prediction = await batcher.submit(query)
             else:
                 supplemented = s.serving.supplement(query)
