@@ -261,6 +261,14 @@ class TestTemplatesOnGPU:
                           Query(user="u1", num=5, black_list=banned))
         assert not set(s.item for s in r2.item_scores) & set(banned)
 
+        # batch_predict: one fused launch for several users must agree
+        # with per-user predict (the micro-batching server path)
+        queries = [(k, Query(user=f"u{k}", num=5)) for k in range(6)]
+        batched = dict(algo.batch_predict(models[0], queries))
+        for k, q in queries:
+            single = algo.predict(models[0], q)
+            assert [s.item for s in batched[k].item_scores] ==                 [s.item for s in single.item_scores]
+
 
 @requires_gpu
 class TestGraphedTopK:
