@@ -37,8 +37,16 @@ def _add_engine_dir(engine_dir: str) -> None:
 
 
 @click.group()
-def cli():
+@click.option("--verbose", is_flag=True,
+              help="INFO-level logging (WorkflowUtils.modifyLogging)")
+@click.option("--debug", is_flag=True, help="DEBUG-level logging")
+def cli(verbose, debug):
     """PredictionIO-AMD — MI355X-native prediction-engine server."""
+    import logging
+    if debug:
+        logging.basicConfig(level=logging.DEBUG)
+    elif verbose:
+        logging.basicConfig(level=logging.INFO)
 
 
 @cli.command()

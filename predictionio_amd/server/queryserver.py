@@ -364,13 +364,39 @@ def create_app(config: ServerConfig,
     return app
 
 
+def undeploy_existing(ip: str, port: int,
+                      access_key: Optional[str] = None) -> bool:
+    """Stop an engine server already bound to ip:port, if any — the
+    reference's deploy replaces the previous server on the port before
+    binding (MasterActor, CreateServer.scala:281-311). Returns True if
+    a server acknowledged the stop."""
+    import urllib.request
+    host = "127.0.0.1" if ip == "0.0.0.0" else ip
+    url = f"http://{host}:{port}/stop"
+    if access_key:
+        url += f"?accessKey={access_key}"
+    try:
+        with urllib.request.urlopen(
+                urllib.request.Request(url, data=b""), timeout=3) as r:
+            if r.status == 200:
+                time.sleep(0.5)  # give it past its SIGTERM timer
+                return True
+    except Exception:
+        pass  # nothing listening (the common case) or not ours
+    return False
+
+
 def run(config: ServerConfig,
         plugins: Optional[List[EngineServerPlugin]] = None,
         ssl_keyfile: Optional[str] = None,
         ssl_certfile: Optional[str] = None) -> None:
     """`pio deploy` entry point (reference default port 8000; SSL via
-    uvicorn per SSLConfiguration.scala)."""
+    uvicorn per SSLConfiguration.scala). Replaces any engine server
+    already on the port first, like the reference's MasterActor."""
     import uvicorn
+    if undeploy_existing(config.ip, config.port, config.access_key):
+        logger.info("replaced a previous engine server on port %s",
+                    config.port)
     uvicorn.run(create_app(config, plugins), host=config.ip,
                 port=config.port, log_level="info",
                 ssl_keyfile=ssl_keyfile, ssl_certfile=ssl_certfile)
