@@ -85,8 +85,25 @@ def status():
 @click.option("--engine-dir", default=".", help="engine template directory")
 def build(engine_dir):
     """Verify the engine directory + compile HIP extensions
-    (commands/Engine.scala:66-165; sbt build replaced by extension build)."""
+    (commands/Engine.scala:66-165; sbt build replaced by extension build).
+    Honors the template.json minimum-version gate
+    (Template.verifyTemplateMinVersion, commands/Template.scala:58)."""
     _add_engine_dir(engine_dir)
+    tj = os.path.join(engine_dir, "template.json")
+    if os.path.exists(tj):
+        import predictionio_amd
+        with open(tj) as f:
+            meta = json.load(f)
+        need = (meta.get("pio", {}).get("version", {}) or {}).get("min")
+        if need:
+            have = tuple(int(x) for x in
+                         predictionio_amd.__version__.split("."))
+            want = tuple(int(x) for x in str(need).split("."))
+            if have < want:
+                click.echo(f"[ERROR] This template requires PIO >= "
+                           f"{need}; installed "
+                           f"{predictionio_amd.__version__}.")
+                raise SystemExit(1)
     from predictionio_amd.ops import build as ops_build
     click.echo("[INFO] building HIP extension (gfx950)...")
     ops_build.build()

@@ -124,3 +124,17 @@ class TestRunCommand:
                                 "a", "b"])
         assert r.exit_code == 0, r.output
         assert fake_engine.RUN_CALLS == [("a", "b")]
+
+
+class TestTemplateMinVersion:
+    def test_build_rejects_newer_min_version(self, tmp_path):
+        import json
+        from click.testing import CliRunner
+        from predictionio_amd.cli.main import cli
+        d = tmp_path / "eng"
+        d.mkdir()
+        (d / "template.json").write_text(
+            json.dumps({"pio": {"version": {"min": "99.0.0"}}}))
+        r = CliRunner().invoke(cli, ["build", "--engine-dir", str(d)])
+        assert r.exit_code == 1
+        assert "requires PIO" in r.output
