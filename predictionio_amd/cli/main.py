@@ -170,8 +170,11 @@ def eval_cmd(evaluation, generator, engine_dir):
               help="coalesce concurrent queries for this window into one "
                    "fused batch_predict launch (0 = per-request)")
 @click.option("--max-batch", default=64)
+@click.option("--ssl-keyfile", default=None)
+@click.option("--ssl-certfile", default=None)
 def deploy(engine_dir, variant, ip, port, feedback, event_server_ip,
-           event_server_port, accesskey, batch_window_ms, max_batch):
+           event_server_port, accesskey, batch_window_ms, max_batch,
+           ssl_keyfile, ssl_certfile):
     """Deploy the latest completed instance behind /queries.json
     (commands/Engine.deploy :208-245 → CreateServer)."""
     _add_engine_dir(engine_dir)
@@ -184,7 +187,7 @@ def deploy(engine_dir, variant, ip, port, feedback, event_server_ip,
         event_server_uri=f"http://{event_server_ip}:{event_server_port}",
         access_key=accesskey, batch_window_ms=batch_window_ms,
         max_batch=max_batch)
-    run(cfg)
+    run(cfg, ssl_keyfile=ssl_keyfile, ssl_certfile=ssl_certfile)
 
 
 @cli.command()
