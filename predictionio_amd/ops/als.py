@@ -80,6 +80,15 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
                     ).contiguous()
         ip, ix, vv = (indptr.contiguous(), indices.contiguous(),
                       values.contiguous())
+        # out-of-range column ids would make the kernels gather OOB from
+        # Y (undefined behavior that can wedge the device) — fail loudly
+        # here instead; one .max() reduction is noise vs the solve
+        if ix.numel():
+            mx = int(ix.max())
+            if mx >= Y.shape[0] or int(ix.min()) < 0:
+                raise ValueError(
+                    f"CSR column id out of range: [{int(ix.min())}, {mx}]"
+                    f" vs {Y.shape[0]} factor rows")
         if implicit and pf <= 128:
             Linv, V = woodbury_lv(Yp, YtYp, lam)
             Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
