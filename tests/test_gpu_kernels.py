@@ -89,6 +89,17 @@ class TestALSKernel:
         assert torch.allclose(X, X_ref, atol=3e-3, rtol=3e-3), \
             f"max abs diff {(X - X_ref).abs().max().item()}"
 
+    def test_oob_column_id_raises(self):
+        """Out-of-range CSR column ids must raise instead of letting the
+        kernel gather OOB (which can wedge the device)."""
+        from predictionio_amd.ops import als as als_ops
+        Y = torch.randn((10, 64)).float().cuda()
+        indptr = torch.tensor([0, 2], dtype=torch.int64).cuda()
+        indices = torch.tensor([1, 10], dtype=torch.int32).cuda()  # 10 OOB
+        values = torch.ones(2).cuda()
+        with pytest.raises(ValueError, match="out of range"):
+            als_ops.als_solve(indptr, indices, values, Y, lam=0.1)
+
     def test_large_row(self):
         """A row with nnz >> chunk size exercises the staging loop."""
         from predictionio_amd.ops import als as als_ops
