@@ -88,6 +88,19 @@ class TestTrainDeployFlow:
         r = runner.invoke(cli, ["export", "--appid", str(app_id),
                                 "--output", str(out)])
         assert r.exit_code == 0 and "Exported 4" in r.output
+        # channel round-trip: import into a named channel, export only it
+        runner.invoke(cli, ["app", "channel-new", "ioapp", "ch1"])
+        r = runner.invoke(cli, ["import", "--appid", str(app_id),
+                                "--channel", "ch1", "--input", str(f)])
+        assert r.exit_code == 0, r.output
+        out2 = tmp_path / "out2.json"
+        r = runner.invoke(cli, ["export", "--appid", str(app_id),
+                                "--channel", "ch1",
+                                "--output", str(out2)])
+        assert r.exit_code == 0 and "Exported 4" in r.output
+        got = [json.loads(l) for l in out2.read_text().splitlines()]
+        assert len(got) == 4
+        assert {e["entityId"] for e in got} == {f"u{i}" for i in range(4)}
 
     def test_template_list(self, runner):
         r = runner.invoke(cli, ["template", "list"])
