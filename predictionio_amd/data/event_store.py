@@ -16,6 +16,8 @@ timeout convention (default 10 s; templates use 200 ms).
 from __future__ import annotations
 
 import concurrent.futures
+import logging
+import threading
 from datetime import datetime
 from typing import Any, Dict, List, Optional
 
@@ -23,7 +25,16 @@ from predictionio_amd.data import storage
 from predictionio_amd.data.events import Event, PropertyMap
 from predictionio_amd.data.storage.base import UNSET
 
-_pool = concurrent.futures.ThreadPoolExecutor(max_workers=8)
+logger = logging.getLogger(__name__)
+
+_POOL_WORKERS = 8
+_pool = concurrent.futures.ThreadPoolExecutor(max_workers=_POOL_WORKERS)
+# in-flight serving lookups (submitted, not yet finished). Timed-out
+# queries are cancelled if still queued; ones already running keep a
+# worker busy until the DB answers, so this gauge is the saturation
+# signal (abandoned queries can pile up under a slow store).
+_inflight = 0
+_inflight_lock = threading.Lock()
 
 
 def app_name_to_id(app_name: str, channel_name: Optional[str] = None):
@@ -98,9 +109,29 @@ def find_by_entity(app_name: str, entity_type: str, entity_id: str,
             start_time=start_time, until_time=until_time,
             limit=limit, reversed=latest))
 
+    global _inflight
+    with _inflight_lock:
+        _inflight += 1
+        inflight = _inflight
+    if inflight > _POOL_WORKERS:
+        logger.warning(
+            "event_store lookup pool saturated: %d in-flight > %d workers "
+            "(slow/locked store backing up serving-time lookups)",
+            inflight, _POOL_WORKERS)
+
+    def _done(_f):
+        global _inflight
+        with _inflight_lock:
+            _inflight -= 1
+
     fut = _pool.submit(_q)
+    fut.add_done_callback(_done)
     try:
         return fut.result(timeout=timeout)
     except concurrent.futures.TimeoutError as e:
+        # if still queued, cancel so abandoned lookups don't consume a
+        # worker; a query already running keeps its worker until the DB
+        # answers (tracked by the _inflight gauge above)
+        fut.cancel()
         raise TimeoutError(
             f"Event store lookup exceeded {timeout}s") from e

@@ -28,23 +28,84 @@ from predictionio_amd.data.storage.base import (
 )
 
 
+class _DoneCursor:
+    """Eagerly-materialized cursor result: rows are fetched while the
+    serializing lock is still held, so no cursor state survives outside
+    the critical section."""
+
+    def __init__(self, cur: sqlite3.Cursor):
+        self.lastrowid = cur.lastrowid
+        self.rowcount = cur.rowcount
+        try:
+            self._rows = cur.fetchall()
+        except sqlite3.ProgrammingError:
+            self._rows = []
+        self._pos = 0
+
+    def fetchone(self):
+        if self._pos < len(self._rows):
+            r = self._rows[self._pos]
+            self._pos += 1
+            return r
+        return None
+
+    def fetchall(self):
+        rows = self._rows[self._pos:]
+        self._pos = len(self._rows)
+        return rows
+
+    def __iter__(self):
+        return iter(self.fetchall())
+
+
+class _SerialConn:
+    """Thread-safe wrapper for the shared :memory: connection: every call
+    holds the lock, and the connection runs in autocommit
+    (isolation_level=None) so statements from different threads can never
+    interleave inside one transaction. DAO code's commit() is then a
+    harmless no-op under the same lock."""
+
+    def __init__(self, conn: sqlite3.Connection, lock: threading.Lock):
+        self._conn = conn
+        self._lock = lock
+
+    def execute(self, *a, **kw):
+        with self._lock:
+            return _DoneCursor(self._conn.execute(*a, **kw))
+
+    def executemany(self, *a, **kw):
+        with self._lock:
+            return _DoneCursor(self._conn.executemany(*a, **kw))
+
+    def commit(self):
+        with self._lock:
+            self._conn.commit()
+
+    def close(self):
+        with self._lock:
+            self._conn.close()
+
+
 class SQLiteClient:
     """One storage source = one sqlite file (or :memory:)."""
 
     def __init__(self, path: str = ":memory:"):
         self.path = path
         self._local = threading.local()
-        self._memory_conn: Optional[sqlite3.Connection] = None
+        self._memory_conn = None
         if path != ":memory:":
             d = os.path.dirname(os.path.abspath(path))
             os.makedirs(d, exist_ok=True)
         else:
-            # :memory: must share one connection across threads
-            self._memory_conn = sqlite3.connect(
-                ":memory:", check_same_thread=False)
+            # :memory: must share one connection across threads; serialize
+            # it (event-server handlers + the event_store pool all hit it)
             self._memory_lock = threading.Lock()
+            self._memory_conn = _SerialConn(
+                sqlite3.connect(":memory:", check_same_thread=False,
+                                isolation_level=None),
+                self._memory_lock)
 
-    def conn(self) -> sqlite3.Connection:
+    def conn(self):
         if self._memory_conn is not None:
             return self._memory_conn
         c = getattr(self._local, "conn", None)

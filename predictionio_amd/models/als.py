@@ -156,10 +156,17 @@ class ALSTrainer:
         self.Y = self._half_step(self.item_csr, self.X, self.n_users)
 
     def fit(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Train and return the FULL (n_users x f, n_items x f) factor
+        matrices. Under multi-GPU training each rank only solves its row
+        blocks, so the local shards are all-gathered at the end — the
+        persisted model must cover the whole catalog (every template saves
+        fit()'s return value against the full user_map/item_map)."""
         if self.X is None:
             self.init_factors()
         for _ in range(self.p.iterations):
             self.step()
+        if pdist.is_distributed():
+            return self.gather_factors()
         return self.X, self.Y
 
     # ------------------------------------------------------------ model out

@@ -48,7 +48,11 @@ def topk_score(Xq: torch.Tensor, Y: torch.Tensor, K: int,
         # phase 2: merge per-slice candidates (small [B, n_slices*K] topk)
         mvals, pos = torch.topk(vals, K, dim=1)
         midx = torch.gather(idxs, 1, pos).long()
-        midx[mvals == float("-inf")] = -1
+        # the kernel pads empty slots with -FLT_MAX (not -inf); normalize
+        # to the CPU reference's -inf / -1 convention for exact parity
+        empty = mvals <= torch.finfo(torch.float32).min
+        midx[empty] = -1
+        mvals = mvals.masked_fill(empty, float("-inf"))
         return mvals, midx
     return topk_score_ref(Xq, Y, K, item_mask, ban_indptr, ban_indices)
 

@@ -165,9 +165,14 @@ class ECommAlgorithm(Algorithm):
         item_map = BiMap.string_int(
             [e.item for e in pd.view_events]
             + [e.item for e in pd.buy_events] + list(pd.items))
-        buy_score = float(self.params.get("buyScore", 1.0))
+        # Reference parity: ECommAlgorithm.genMLlibRating builds MLlib
+        # ratings from viewEvents ONLY (ECommAlgorithm.scala:168-205);
+        # buys feed just the popularity counts. buyScore>0 blends buys
+        # into training as an opt-in extension beyond the reference.
+        buy_score = float(self.params.get("buyScore", 0.0))
         evs = ([(e, 1.0) for e in pd.view_events]
-               + [(e, buy_score) for e in pd.buy_events])
+               + ([(e, buy_score) for e in pd.buy_events]
+                  if buy_score > 0.0 else []))
         users = torch.tensor([user_map[e.user] for e, _ in evs],
                              dtype=torch.int32)
         items = torch.tensor([item_map[e.item] for e, _ in evs],
