@@ -15,7 +15,8 @@ import sysconfig
 OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(OPS_DIR, "csrc")
 SO_PATH = os.path.join(OPS_DIR, "_pio_hip.so")
-SOURCES = ["bindings.cpp", "als_kernels.hip", "topk_kernels.hip"]
+SOURCES = ["bindings.cpp", "als_kernels.hip", "topk_kernels.hip",
+           "topk_mfma.hip"]
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 

@@ -24,6 +24,13 @@ extern "C" void launch_topk_score(
     int B, long long N, int f, int K, int n_slices, int item_base,
     unsigned long long* prof, hipStream_t stream);
 
+extern "C" void launch_topk_mfma(
+    const unsigned short* Xq, const unsigned short* Y,
+    const uint8_t* item_mask, const long long* ban_indptr,
+    const int* ban_indices, float* out_val, int* out_idx,
+    int B, long long N, int f, int K, int n_slices, int item_base,
+    unsigned long long* prof, hipStream_t stream);
+
 namespace {
 
 void check_cuda_f32(const torch::Tensor& t, const char* name) {
@@ -163,6 +170,76 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score(
   return {out_val, out_idx};
 }
 
+// MFMA variant of topk_score: bf16 query/item factors (fp32 accumulate on
+// the matrix cores). Same output layout as topk_score — candidate groups
+// (n_slices*4) per query; the Python wrapper merges and fp32-rescores.
+std::tuple<torch::Tensor, torch::Tensor> topk_score_mfma(
+    torch::Tensor Xq, torch::Tensor Y, int64_t K, int64_t n_slices,
+    c10::optional<torch::Tensor> item_mask,
+    c10::optional<torch::Tensor> ban_indptr,
+    c10::optional<torch::Tensor> ban_indices, int64_t item_base,
+    c10::optional<torch::Tensor> prof) {
+  TORCH_CHECK(Xq.is_cuda() && Xq.is_contiguous() &&
+                  Xq.scalar_type() == torch::kBFloat16,
+              "Xq must be contiguous bf16 GPU");
+  TORCH_CHECK(Y.is_cuda() && Y.is_contiguous() &&
+                  Y.scalar_type() == torch::kBFloat16,
+              "Y must be contiguous bf16 GPU");
+  const int64_t f = Y.size(1);
+  TORCH_CHECK(f == 32 || f == 64 || f == 128,
+              "mfma rank must be one of 32/64/128, got ", f);
+  TORCH_CHECK(Xq.size(1) == f, "Xq/Y rank mismatch");
+  TORCH_CHECK(K >= 1 && K <= 64, "K must be in [1, 64]");
+  const int64_t B = Xq.size(0);
+  const int64_t N = Y.size(0);
+  TORCH_CHECK(n_slices >= 1);
+  const uint8_t* mask_ptr = nullptr;
+  if (item_mask.has_value()) {
+    TORCH_CHECK(item_mask->is_cuda() && item_mask->is_contiguous() &&
+                    item_mask->scalar_type() == torch::kUInt8,
+                "item_mask must be contiguous u8 GPU");
+    TORCH_CHECK(item_mask->size(0) == N, "item_mask size mismatch");
+    mask_ptr = item_mask->data_ptr<uint8_t>();
+  }
+  const long long* bi_ptr = nullptr;
+  const int* bx_ptr = nullptr;
+  if (ban_indptr.has_value()) {
+    TORCH_CHECK(ban_indices.has_value(), "ban_indices required");
+    TORCH_CHECK(ban_indptr->is_cuda() && ban_indptr->is_contiguous() &&
+                ban_indptr->scalar_type() == torch::kInt64);
+    TORCH_CHECK(ban_indices->is_cuda() && ban_indices->is_contiguous() &&
+                ban_indices->scalar_type() == torch::kInt32);
+    TORCH_CHECK(ban_indptr->size(0) == B + 1, "ban_indptr must be B+1");
+    bi_ptr = reinterpret_cast<const long long*>(
+        ban_indptr->data_ptr<int64_t>());
+    bx_ptr = ban_indices->data_ptr<int>();
+  }
+  auto out_val = torch::empty({B, n_slices * 4 * K},
+                              Xq.options().dtype(torch::kFloat32));
+  auto out_idx = torch::empty({B, n_slices * 4 * K},
+                              Xq.options().dtype(torch::kInt32));
+  unsigned long long* prof_ptr = nullptr;
+  if (prof.has_value()) {
+    TORCH_CHECK(prof->is_cuda() && prof->is_contiguous() &&
+                    prof->scalar_type() == torch::kUInt64 &&
+                    prof->numel() >= 5,
+                "prof must be a contiguous u64[5] GPU tensor");
+    prof_ptr = reinterpret_cast<unsigned long long*>(
+        prof->data_ptr<uint64_t>());
+  }
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
+  hipStream_t stream =
+      c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  launch_topk_mfma(
+      reinterpret_cast<const unsigned short*>(Xq.data_ptr()),
+      reinterpret_cast<const unsigned short*>(Y.data_ptr()), mask_ptr,
+      bi_ptr, bx_ptr, out_val.data_ptr<float>(), out_idx.data_ptr<int>(),
+      (int)B, (long long)N, (int)f, (int)K, (int)n_slices, (int)item_base,
+      prof_ptr, stream);
+  C10_HIP_CHECK(hipGetLastError());
+  return {out_val, out_idx};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -175,6 +252,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("wr_scale") = true, py::arg("which") = 0,
         py::arg("out") = py::none(), py::arg("prof") = py::none());
   m.def("topk_score", &topk_score, "Fused masked top-K scoring",
+        py::arg("Xq"), py::arg("Y"), py::arg("K"), py::arg("n_slices") = 64,
+        py::arg("item_mask") = py::none(), py::arg("ban_indptr") = py::none(),
+        py::arg("ban_indices") = py::none(), py::arg("item_base") = 0,
+        py::arg("prof") = py::none());
+  m.def("topk_score_mfma", &topk_score_mfma,
+        "Fused masked top-K scoring on MFMA (bf16 in, fp32 accumulate)",
         py::arg("Xq"), py::arg("Y"), py::arg("K"), py::arg("n_slices") = 64,
         py::arg("item_mask") = py::none(), py::arg("ban_indptr") = py::none(),
         py::arg("ban_indices") = py::none(), py::arg("item_base") = 0,

@@ -1,0 +1,300 @@
+// MFMA-based masked top-K scoring for MI355X (gfx950, CDNA4).
+//
+// v5 of the serve hot path: the B x N x F score computation runs on the
+// matrix cores (v_mfma_f32_16x16x32_bf16, bf16 inputs / fp32 accumulate)
+// instead of the VALU fp32 dot loop of topk_score_kernel (v3). Replaces
+// the same reference semantics (recommendProductsWithFilter
+// examples/.../ALSModel.scala:44-60, ecommerce predictKnownUser
+// ECommAlgorithm.scala:471-506, similarproduct cosine top-N
+// ALSAlgorithm.scala:168-242); the Python wrapper re-checks the bf16
+// survivors against the fp32 factors so served scores stay exact.
+//
+// Geometry (per 256-thread workgroup, 4 waves):
+//  - the block owns 64 queries; wave w owns queries [16w, 16w+16)
+//  - item slices as in v3: grid = (ublocks, n_slices); blockIdx.x is the
+//    ublock so concurrently-resident workgroups share one Y slice
+//    (~N/n_slices rows) through the 256 MB L3
+//  - per chunk of TM_CHUNK=64 items: Y rows are reg-staged into LDS as
+//    bf16 with an XOR swizzle ((row & SWM) << 4 on the byte offset, the
+//    T2 bank-conflict fix for "different rows, same col-range" b128
+//    reads), then each wave runs (TM_CHUNK/16) x (F/32) MFMAs:
+//       A = Y tile   (lane holds Y[item = l&15][k = (l>>4)*8 + i])
+//       B = X^T tile (lane holds X[query = l&15][k = (l>>4)*8 + i]),
+//    accumulating D[item][query] in fp32 (D: col = l&15 = query,
+//    row = (l>>4)*4 + reg = item) — operand maps per the CK xdlops
+//    contract and the measured gfx950 C/D layout.
+//  - epilogue runs on the accumulator registers directly: lane l keeps a
+//    private top-K list for query (l&15) over the items of its lane
+//    group (l>>4) — per query there are 4 lists per slice, written out
+//    as candidate group (slice*4 + l>>4), identical output layout to v3
+//    so the host merge is unchanged.
+//
+// The X fragments live in registers for the whole kernel (loaded once);
+// item staging is software-pipelined through registers like v3 so the
+// next chunk's global loads fly during the MFMA phase.
+
+#include <float.h>
+#include <hip/hip_runtime.h>
+
+#define TM_CHUNK 64   // items staged per LDS pass
+#define TM_WAVES 4
+#define TM_QPW 16     // queries per wave
+#define TM_UPB (TM_WAVES * TM_QPW)  // queries per block
+
+typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+__device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
+  int lo = 0, hi = n - 1;
+  while (lo <= hi) {
+    int mid = (lo + hi) >> 1;
+    int v = arr[mid];
+    if (v == x) return true;
+    if (v < x) lo = mid + 1; else hi = mid - 1;
+  }
+  return false;
+}
+
+template <int F, bool PROF>
+__global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
+    const unsigned short* __restrict__ Xq,   // B x F bf16
+    const unsigned short* __restrict__ Y,    // N x F bf16
+    const uint8_t* __restrict__ item_mask,   // N or nullptr
+    const long long* __restrict__ ban_indptr,
+    const int* __restrict__ ban_indices,
+    float* __restrict__ out_val,             // B x (n_slices*4) x K
+    int* __restrict__ out_idx,
+    int B, long long N, int K, int n_slices, int item_base,
+    unsigned long long* prof)
+{
+  constexpr int ROWB = F * 2;            // bytes per staged Y row
+  constexpr int SWM = (F >= 64) ? 7 : 3; // XOR-swizzle row mask
+  constexpr int KS = F / 32;             // MFMA K-steps per dot product
+  constexpr int IFR = TM_CHUNK / 16;     // item fragments per chunk
+  extern __shared__ char lds_raw[];
+  unsigned short* ys = reinterpret_cast<unsigned short*>(lds_raw);
+  float* topv = reinterpret_cast<float*>(lds_raw + TM_CHUNK * ROWB);
+  const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
+  int* topi = reinterpret_cast<int*>(topv + 256 * KP);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int lg = lane >> 4;   // lane group: k-slice on input, item rows out
+  const int lq = lane & 15;   // query column (B operand / D col)
+
+  const bool probe = PROF && tid == 0;
+  unsigned long long pt = 0, acc_setup = 0, acc_stage = 0, acc_score = 0;
+  if (probe) pt = wall_clock64();
+
+  const long long u0 = (long long)blockIdx.x * TM_UPB;
+  const long long guser = u0 + wave * TM_QPW + lq;
+  const bool has_user = guser < B;
+  const int slice = blockIdx.y;
+  const long long per = (N + n_slices - 1) / n_slices;
+  const long long it0 = (long long)slice * per;
+  const long long it1 = min(N, it0 + per);
+
+  // ---- X fragments: one b128 per K-step, held for the whole kernel
+  bf16x8 xf[KS];
+#pragma unroll
+  for (int ks = 0; ks < KS; ++ks) {
+    if (has_user) {
+      xf[ks] = *reinterpret_cast<const bf16x8*>(
+          &Xq[guser * F + ks * 32 + lg * 8]);
+    } else {
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      xf[ks] = z;
+    }
+  }
+
+  // ---- init per-lane top-K lists
+  for (int e = tid; e < 256 * KP; e += 256) {
+    topv[e] = -FLT_MAX;
+    topi[e] = -1;
+  }
+  __syncthreads();
+  float th = -FLT_MAX;
+  float* tvu = topv + tid * KP;
+  int* tiu = topi + tid * KP;
+
+  const int* ban = nullptr;
+  int bn = 0;
+  if (ban_indptr != nullptr && has_user) {
+    const long long b0 = ban_indptr[guser];
+    bn = (int)(ban_indptr[guser + 1] - b0);
+    ban = ban_indices + b0;
+  }
+  if (probe) {
+    const unsigned long long now = wall_clock64();
+    acc_setup = now - pt;
+    pt = now;
+  }
+
+  // ---- software-pipelined staging registers (granules of 16 B)
+  constexpr int NG = (TM_CHUNK * ROWB) / 16 / 256;  // granules per thread
+  static_assert(NG >= 1, "chunk must cover one granule per thread");
+  u32x4 stg[NG];
+  {
+    const long long base = it0;
+#pragma unroll
+    for (int r = 0; r < NG; ++r) {
+      const int lin = (tid + r * 256) * 16;
+      const int row = lin / ROWB;
+      const int col = lin % ROWB;
+      stg[r] = (base + row < it1)
+          ? *reinterpret_cast<const u32x4*>(&Y[(base + row) * F + col / 2])
+          : u32x4{0u, 0u, 0u, 0u};
+    }
+  }
+
+  for (long long base = it0; base < it1; base += TM_CHUNK) {
+    __syncthreads();  // all waves done reading ys from the previous chunk
+    // drain the in-flight registers into LDS (swizzled), then issue the
+    // next chunk's loads so they fly during the MFMA phase
+#pragma unroll
+    for (int r = 0; r < NG; ++r) {
+      const int lin = (tid + r * 256) * 16;
+      const int row = lin / ROWB;
+      const int col = lin % ROWB;
+      const int dst = row * ROWB + (col ^ ((row & SWM) << 4));
+      *reinterpret_cast<u32x4*>(reinterpret_cast<char*>(ys) + dst) = stg[r];
+    }
+    const long long nbase = base + TM_CHUNK;
+    if (nbase < it1) {
+#pragma unroll
+      for (int r = 0; r < NG; ++r) {
+        const int lin = (tid + r * 256) * 16;
+        const int row = lin / ROWB;
+        const int col = lin % ROWB;
+        stg[r] = (nbase + row < it1)
+            ? *reinterpret_cast<const u32x4*>(
+                  &Y[(nbase + row) * F + col / 2])
+            : u32x4{0u, 0u, 0u, 0u};
+      }
+    }
+    __syncthreads();
+    if (probe) {
+      const unsigned long long now = wall_clock64();
+      acc_stage += now - pt;
+      pt = now;
+    }
+
+    // ---- MFMA: every wave scores the whole chunk for its 16 queries
+    f32x4 acc[IFR];
+#pragma unroll
+    for (int i = 0; i < IFR; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+#pragma unroll
+      for (int i = 0; i < IFR; ++i) {
+        const int row = i * 16 + lq;           // A operand row = item
+        const int col = ks * 64 + lg * 16;     // byte offset of k-slice
+        const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const char*>(ys) +
+            row * ROWB + (col ^ ((row & SWM) << 4)));
+        acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, xf[ks], acc[i], 0, 0, 0);
+      }
+    }
+
+    // ---- epilogue straight off the accumulators
+#pragma unroll
+    for (int i = 0; i < IFR; ++i) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long long item = base + i * 16 + lg * 4 + r;  // D row
+        const float s = acc[i][r];
+        if (has_user && item < it1 && s > th) {
+          if ((item_mask == nullptr || !item_mask[item]) &&
+              (ban == nullptr ||
+               !tm_in_sorted(ban, bn, (int)(item + item_base)))) {
+            int mi = 0;
+            float mv = tvu[0];
+            for (int q = 1; q < K; ++q)
+              if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
+            tvu[mi] = s;
+            tiu[mi] = (int)(item + item_base);
+            float nm = tvu[0];
+            for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
+            th = nm;
+          }
+        }
+      }
+    }
+    if (probe) {
+      const unsigned long long now = wall_clock64();
+      acc_score += now - pt;
+      pt = now;
+    }
+  }
+  __syncthreads();
+  if (probe) pt = wall_clock64();
+
+  // ---- write out: list (wave w, lane l) serves query u0+16w+(l&15),
+  // candidate group slice*4 + (l>>4) — same output layout as v3
+  for (int e = tid; e < 256 * K; e += 256) {
+    const int list = e / K;
+    const int q = e % K;
+    const int w = list >> 6;
+    const int l = list & 63;
+    const long long gu = u0 + w * TM_QPW + (l & 15);
+    if (gu < B) {
+      const long long g = (long long)slice * TM_WAVES + (l >> 4);
+      const long long o = (gu * n_slices * TM_WAVES + g) * K + q;
+      out_val[o] = topv[list * KP + q];
+      out_idx[o] = topi[list * KP + q];
+    }
+  }
+  if (probe) {
+    atomicAdd(&prof[0], acc_setup);
+    atomicAdd(&prof[1], acc_stage);
+    atomicAdd(&prof[2], acc_score);
+    atomicAdd(&prof[3], wall_clock64() - pt);
+    atomicAdd(&prof[4], 1ull);
+  }
+}
+
+extern "C" void launch_topk_mfma(
+    const unsigned short* Xq, const unsigned short* Y,
+    const uint8_t* item_mask, const long long* ban_indptr,
+    const int* ban_indices, float* out_val, int* out_idx,
+    int B, long long N, int f, int K, int n_slices, int item_base,
+    unsigned long long* prof, hipStream_t stream)
+{
+  dim3 grid((B + TM_UPB - 1) / TM_UPB, n_slices);
+  dim3 block(256);
+#define LAUNCH_M(FF)                                                         \
+  do {                                                                       \
+    size_t lds_bytes = (size_t)TM_CHUNK * (FF * 2) +                         \
+                       (sizeof(float) + sizeof(int)) * 256 * (K + 1);        \
+    static bool attr_set_##FF = false;                                       \
+    if (!attr_set_##FF && lds_bytes > 64 * 1024) {                           \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, false>),       \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, true>),        \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
+      attr_set_##FF = true;                                                  \
+    }                                                                        \
+    if (prof != nullptr)                                                     \
+      hipLaunchKernelGGL((topk_mfma_kernel<FF, true>), grid, block,          \
+                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
+                         ban_indices, out_val, out_idx, B, N, K, n_slices,   \
+                         item_base, prof);                                   \
+    else                                                                     \
+      hipLaunchKernelGGL((topk_mfma_kernel<FF, false>), grid, block,         \
+                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
+                         ban_indices, out_val, out_idx, B, N, K, n_slices,   \
+                         item_base, nullptr);                                \
+  } while (0)
+  switch (f) {
+    case 32: LAUNCH_M(32); break;
+    case 64: LAUNCH_M(64); break;
+    case 128: LAUNCH_M(128); break;
+    default: break;
+  }
+#undef LAUNCH_M
+}

@@ -118,16 +118,18 @@ class TestALSKernel:
 
 @requires_gpu
 class TestTopKKernel:
+    @pytest.mark.parametrize("mode", ["fp32", "mfma"])
     @pytest.mark.parametrize("f", [32, 64])
     @pytest.mark.parametrize("K", [1, 4, 20])
-    def test_matches_reference(self, f, K):
+    def test_matches_reference(self, f, K, mode):
         from predictionio_amd.ops import topk as topk_ops
         g = torch.Generator().manual_seed(f * 100 + K)
         B, N = 37, 5000
         Xq = torch.randn((B, f), generator=g).float()
         Y = torch.randn((N, f), generator=g).float()
         rv, ri = topk_ops.topk_score_ref(Xq, Y, K)
-        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K, n_slices=7)
+        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K, n_slices=7,
+                                     mode=mode)
         gv, gi = gv.cpu(), gi.cpu()
         assert torch.allclose(gv, rv, atol=1e-4, rtol=1e-4), \
             f"max val diff {(gv - rv).abs().max()}"
@@ -135,7 +137,8 @@ class TestTopKKernel:
         chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
         assert torch.allclose(chosen, rv, atol=1e-4, rtol=1e-4)
 
-    def test_masks(self):
+    @pytest.mark.parametrize("mode", ["fp32", "mfma"])
+    def test_masks(self, mode):
         from predictionio_amd.ops import topk as topk_ops
         g = torch.Generator().manual_seed(5)
         B, N, f, K = 16, 3000, 64, 10
@@ -152,7 +155,8 @@ class TestTopKKernel:
         gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K,
                                      item_mask=mask.cuda(),
                                      ban_indptr=bi.cuda(),
-                                     ban_indices=bx.cuda(), n_slices=5)
+                                     ban_indices=bx.cuda(), n_slices=5,
+                                     mode=mode)
         gv, gi = gv.cpu(), gi.cpu()
         assert torch.allclose(gv, rv, atol=1e-4, rtol=1e-4)
         # no banned index may appear
@@ -160,6 +164,41 @@ class TestTopKKernel:
             banned = set(bans[b].tolist()) | set(
                 torch.nonzero(mask).flatten().tolist())
             assert not (set(gi[b].tolist()) - {-1}) & banned
+
+    @pytest.mark.parametrize("f", [32, 64, 128])
+    def test_mfma_matches_fp32_reference(self, f):
+        """MFMA path (bf16 score + fp32 rescore) vs the fp32 torch
+        reference: top-K values must match exactly-ish (returned values
+        ARE fp32 dots of the chosen items) and recall must be ~perfect
+        on random data — the 2x candidate margin covers bf16 rounding."""
+        from predictionio_amd.ops import topk as topk_ops
+        g = torch.Generator().manual_seed(f)
+        B, N, K = 130, 20000, 20
+        Xq = torch.randn((B, f), generator=g).float()
+        Y = torch.randn((N, f), generator=g).float()
+        rv, ri = topk_ops.topk_score_ref(Xq, Y, K)
+        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K, mode="mfma")
+        gv, gi = gv.cpu(), gi.cpu()
+        # recall@K vs the fp32 reference
+        hits = sum(len(set(gi[b].tolist()) & set(ri[b].tolist()))
+                   for b in range(B))
+        recall = hits / (B * K)
+        assert recall >= 0.999, f"recall {recall}"
+        # returned values are exact fp32 scores of the returned items
+        chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
+        assert torch.allclose(gv, chosen, atol=1e-5, rtol=1e-5)
+        # and the K-th value can differ from the reference only by a
+        # bf16-near-tie at the candidate cut
+        assert torch.allclose(gv, rv, atol=5e-3, rtol=5e-3)
+
+    def test_mfma_more_k_than_items(self):
+        from predictionio_amd.ops import topk as topk_ops
+        Xq = torch.randn((3, 64)).float().cuda()
+        Y = torch.randn((10, 64)).float().cuda()
+        gv, gi = topk_ops.topk_score(Xq, Y, 20, n_slices=2, mode="mfma")
+        assert gi.shape == (3, 20)
+        assert (gi[:, 10:] == -1).all()
+        assert (gv[:, 10:] == float("-inf")).all()
 
     def test_small_k_full_block(self):
         """K small enough that ys+lists < the 17.4 KB query-staging pass,
