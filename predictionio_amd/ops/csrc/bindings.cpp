@@ -214,9 +214,11 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score_mfma(
         ban_indptr->data_ptr<int64_t>());
     bx_ptr = ban_indices->data_ptr<int>();
   }
-  auto out_val = torch::empty({B, n_slices * 4 * K},
+  // one candidate group per slice per query (the kernel keeps a single
+  // shared top-K list per query per workgroup)
+  auto out_val = torch::empty({B, n_slices * K},
                               Xq.options().dtype(torch::kFloat32));
-  auto out_idx = torch::empty({B, n_slices * 4 * K},
+  auto out_idx = torch::empty({B, n_slices * K},
                               Xq.options().dtype(torch::kInt32));
   unsigned long long* prof_ptr = nullptr;
   if (prof.has_value()) {

@@ -198,6 +198,21 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
           dst[3] = stg4[r].w;
         }
       }
+    } else {
+      // coalesced stage of cn item rows
+      for (int e = tid; e < cn * F; e += 256) {
+        const int c = e / F;
+        const int k = e % F;
+        ys[c * FP + k] = Y[(base + c) * F + k];
+      }
+    }
+    __syncthreads();
+    if (PIPE) {
+      // issue the NEXT chunk's loads AFTER the barrier: __syncthreads
+      // compiles to s_waitcnt vmcnt(0), so loads issued before it were
+      // drained at the barrier (the phase probe's "stage = inter-wave
+      // barrier wait" signature). Here they fly under the score loop
+      // and are awaited by the next iteration's first barrier.
       const long long nbase = base + TK_CHUNK;
       if (nbase < it1) {
         const int cnn = (int)min((long long)TK_CHUNK, it1 - nbase);
@@ -212,15 +227,7 @@ __global__ __launch_bounds__(256, 3) void topk_score_kernel(
               : f32x4{0.f, 0.f, 0.f, 0.f};
         }
       }
-    } else {
-      // coalesced stage of cn item rows
-      for (int e = tid; e < cn * F; e += 256) {
-        const int c = e / F;
-        const int k = e % F;
-        ys[c * FP + k] = Y[(base + c) * F + k];
-      }
     }
-    __syncthreads();
     if (probe) {
       const unsigned long long now = wall_clock64();
       acc_stage += now - pt;

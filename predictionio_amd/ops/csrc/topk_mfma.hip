@@ -11,9 +11,8 @@
 //
 // Geometry (per 256-thread workgroup, 4 waves):
 //  - the block owns 64 queries; wave w owns queries [16w, 16w+16)
-//  - item slices as in v3: grid = (ublocks, n_slices); blockIdx.x is the
-//    ublock so concurrently-resident workgroups share one Y slice
-//    (~N/n_slices rows) through the 256 MB L3
+//  - item slices: grid = (ublocks, n_slices); blockIdx.x is the ublock
+//    so concurrently-resident workgroups share one Y slice through L3
 //  - per chunk of TM_CHUNK=64 items: Y rows are reg-staged into LDS as
 //    bf16 with an XOR swizzle ((row & SWM) << 4 on the byte offset, the
 //    T2 bank-conflict fix for "different rows, same col-range" b128
@@ -23,15 +22,32 @@
 //    accumulating D[item][query] in fp32 (D: col = l&15 = query,
 //    row = (l>>4)*4 + reg = item) — operand maps per the CK xdlops
 //    contract and the measured gfx950 C/D layout.
-//  - epilogue runs on the accumulator registers directly: lane l keeps a
-//    private top-K list for query (l&15) over the items of its lane
-//    group (l>>4) — per query there are 4 lists per slice, written out
-//    as candidate group (slice*4 + l>>4), identical output layout to v3
-//    so the host merge is unchanged.
+//
+// Top-K epilogue (v2 — the round-2 phase probe showed the v1 per-lane
+// list scheme spending 63% of kernel time on insert machinery):
+//  - ONE top-K list per (query, workgroup) in LDS, with the running
+//    K-th-best threshold also in LDS (th_lds[wave*16 + query]).
+//  - common case per chunk: each lane max-reduces its 16 accumulator
+//    values (15 v_max), reads the shared threshold (one broadcast
+//    ds_read) and skips everything else — ~25 instructions per
+//    64 items x 16 queries.
+//  - rare case (some lane's max beats the threshold): the 4 lanes of a
+//    query's quad take turns (serialized by lane group, exec-masked —
+//    a single wave executes groups in program order, and within one
+//    group every active lane owns a DIFFERENT query, so list writes
+//    never contend) scanning their values and inserting; masks/bans are
+//    only consulted here.
+//  - a shared per-query threshold is strictly tighter than v1's
+//    per-lane-group thresholds, cutting total inserts ~4x (inserts per
+//    query = K ln(items/K) per LIST, and lists per query dropped 4x),
+//    and the list LDS shrinks 4x (64 lists instead of 256), lifting
+//    occupancy from 3 to 7 workgroups/CU.
 //
 // The X fragments live in registers for the whole kernel (loaded once);
-// item staging is software-pipelined through registers like v3 so the
-// next chunk's global loads fly during the MFMA phase.
+// item staging is software-pipelined through registers, with the next
+// chunk's loads issued AFTER the second barrier so they fly under the
+// MFMA phase (a __syncthreads compiles to s_waitcnt vmcnt(0) — loads
+// issued before it would be drained at the barrier).
 
 #include <float.h>
 #include <hip/hip_runtime.h>
@@ -57,13 +73,13 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
 }
 
 template <int F, bool PROF>
-__global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
+__global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
     const uint8_t* __restrict__ item_mask,   // N or nullptr
     const long long* __restrict__ ban_indptr,
     const int* __restrict__ ban_indices,
-    float* __restrict__ out_val,             // B x (n_slices*4) x K
+    float* __restrict__ out_val,             // B x n_slices x K
     int* __restrict__ out_idx,
     int B, long long N, int K, int n_slices, int item_base,
     unsigned long long* prof)
@@ -76,7 +92,8 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
   unsigned short* ys = reinterpret_cast<unsigned short*>(lds_raw);
   float* topv = reinterpret_cast<float*>(lds_raw + TM_CHUNK * ROWB);
   const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
-  int* topi = reinterpret_cast<int*>(topv + 256 * KP);
+  int* topi = reinterpret_cast<int*>(topv + TM_UPB * KP);
+  float* th_lds = reinterpret_cast<float*>(topi + TM_UPB * KP);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -109,15 +126,16 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
     }
   }
 
-  // ---- init per-lane top-K lists
-  for (int e = tid; e < 256 * KP; e += 256) {
+  // ---- init the per-query top-K lists + shared thresholds
+  for (int e = tid; e < TM_UPB * KP; e += 256) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
+  if (tid < TM_UPB) th_lds[tid] = -FLT_MAX;
   __syncthreads();
-  float th = -FLT_MAX;
-  float* tvu = topv + tid * KP;
-  int* tiu = topi + tid * KP;
+  const int mylist = wave * TM_QPW + lq;   // this lane's query list
+  float* tvu = topv + mylist * KP;
+  int* tiu = topi + mylist * KP;
 
   const int* ban = nullptr;
   int bn = 0;
@@ -151,8 +169,7 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
 
   for (long long base = it0; base < it1; base += TM_CHUNK) {
     __syncthreads();  // all waves done reading ys from the previous chunk
-    // drain the in-flight registers into LDS (swizzled), then issue the
-    // next chunk's loads so they fly during the MFMA phase
+    // drain the in-flight registers into LDS (swizzled)
 #pragma unroll
     for (int r = 0; r < NG; ++r) {
       const int lin = (tid + r * 256) * 16;
@@ -161,6 +178,10 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
       const int dst = row * ROWB + (col ^ ((row & SWM) << 4));
       *reinterpret_cast<u32x4*>(reinterpret_cast<char*>(ys) + dst) = stg[r];
     }
+    __syncthreads();
+    // issue the NEXT chunk's global loads here — after the barrier —
+    // so they fly under the MFMA phase and are only awaited by the
+    // next iteration's first barrier
     const long long nbase = base + TM_CHUNK;
     if (nbase < it1) {
 #pragma unroll
@@ -174,7 +195,6 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
             : u32x4{0u, 0u, 0u, 0u};
       }
     }
-    __syncthreads();
     if (probe) {
       const unsigned long long now = wall_clock64();
       acc_stage += now - pt;
@@ -199,26 +219,51 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
       }
     }
 
-    // ---- epilogue straight off the accumulators
+    // ---- epilogue: common case is one max-reduce + one threshold read
+    float mymax = -FLT_MAX;
 #pragma unroll
-    for (int i = 0; i < IFR; ++i) {
+    for (int i = 0; i < IFR; ++i)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const long long item = base + i * 16 + lg * 4 + r;  // D row
-        const float s = acc[i][r];
-        if (has_user && item < it1 && s > th) {
-          if ((item_mask == nullptr || !item_mask[item]) &&
-              (ban == nullptr ||
-               !tm_in_sorted(ban, bn, (int)(item + item_base)))) {
-            int mi = 0;
-            float mv = tvu[0];
-            for (int q = 1; q < K; ++q)
-              if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
-            tvu[mi] = s;
-            tiu[mi] = (int)(item + item_base);
-            float nm = tvu[0];
-            for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
-            th = nm;
+      for (int r = 0; r < 4; ++r) mymax = fmaxf(mymax, acc[i][r]);
+    const float th_b = th_lds[mylist];
+    if (has_user && mymax > th_b) {
+      // serialize the query's 4 lane groups: exec-masked blocks of one
+      // wave run in program order, and within a group the active lanes
+      // all own different queries, so list writes never contend.
+      // MUST be a runtime loop with an opaque barrier per iteration:
+      // with `#pragma unroll` the four structurally-identical blocks
+      // got tail-merged by the compiler into ONE exec-masked block
+      // (it does not model cross-lane LDS aliasing), so all groups
+      // read the threshold before any insert — losing inserts (the
+      // round-2 K=1 test failure: later groups overwrote the max).
+#pragma unroll 1
+      for (int g = 0; g < 4; ++g) {
+        asm volatile("" ::: "memory");  // keep iterations distinct
+        if (lg == g) {
+          float th = th_lds[mylist];
+#pragma unroll
+          for (int i = 0; i < IFR; ++i) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const long long item = base + i * 16 + lg * 4 + r;  // D row
+              const float s = acc[i][r];
+              if (s > th && item < it1) {
+                if ((item_mask == nullptr || !item_mask[item]) &&
+                    (ban == nullptr ||
+                     !tm_in_sorted(ban, bn, (int)(item + item_base)))) {
+                  int mi = 0;
+                  float mv = tvu[0];
+                  for (int q = 1; q < K; ++q)
+                    if (tvu[q] < mv) { mv = tvu[q]; mi = q; }
+                  tvu[mi] = s;
+                  tiu[mi] = (int)(item + item_base);
+                  float nm = tvu[0];
+                  for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
+                  th = nm;
+                  th_lds[mylist] = nm;
+                }
+              }
+            }
           }
         }
       }
@@ -232,17 +277,13 @@ __global__ __launch_bounds__(256, 3) void topk_mfma_kernel(
   __syncthreads();
   if (probe) pt = wall_clock64();
 
-  // ---- write out: list (wave w, lane l) serves query u0+16w+(l&15),
-  // candidate group slice*4 + (l>>4) — same output layout as v3
-  for (int e = tid; e < 256 * K; e += 256) {
+  // ---- write out: one candidate group per slice per query
+  for (int e = tid; e < TM_UPB * K; e += 256) {
     const int list = e / K;
     const int q = e % K;
-    const int w = list >> 6;
-    const int l = list & 63;
-    const long long gu = u0 + w * TM_QPW + (l & 15);
+    const long long gu = u0 + list;
     if (gu < B) {
-      const long long g = (long long)slice * TM_WAVES + (l >> 4);
-      const long long o = (gu * n_slices * TM_WAVES + g) * K + q;
+      const long long o = (gu * n_slices + slice) * (long long)K + q;
       out_val[o] = topv[list * KP + q];
       out_idx[o] = topi[list * KP + q];
     }
@@ -268,7 +309,8 @@ extern "C" void launch_topk_mfma(
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
     size_t lds_bytes = (size_t)TM_CHUNK * (FF * 2) +                         \
-                       (sizeof(float) + sizeof(int)) * 256 * (K + 1);        \
+                       (sizeof(float) + sizeof(int)) * TM_UPB * (K + 1) +    \
+                       sizeof(float) * TM_UPB;                               \
     static bool attr_set_##FF = false;                                       \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                           \
       hipFuncSetAttribute(                                                   \

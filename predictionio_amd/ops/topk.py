@@ -117,8 +117,11 @@ def _topk_score_mfma(Xq: torch.Tensor, Y: torch.Tensor, K: int,
     pf = _mfma_rank(f)
     Xp = Xq if pf == f else torch.nn.functional.pad(Xq, (0, pf - f))
     if n_slices is None:
+        # insert volume grows ~linearly with n_slices (each slice resets
+        # the per-query threshold), so use just enough slices to fill
+        # the chip (measured sweep: scripts/mfma_phase_probe.py)
         ublocks = (B + 63) // 64
-        n_slices = max(1, min(2048 // ublocks + 1, (N + 255) // 256))
+        n_slices = max(2, min(1536 // ublocks, (N + 255) // 256))
     # Y's bf16 copy is cached (factors are static across serving batches);
     # Xq is cast per call — it changes every batch, and under hipGraph
     # capture (GraphedTopK) the cast must be part of the captured work
