@@ -64,6 +64,7 @@ class ALSTrainer:
         self.item_csr = None   # (indptr, indices, values) local items x users
         self.X: Optional[torch.Tensor] = None  # local user factors
         self.Y: Optional[torch.Tensor] = None  # local item factors
+        self._pending_y = None  # in-flight Y gather from the item half
         self.phase_times = {"gather_s": 0.0, "solve_s": 0.0}
 
     # ------------------------------------------------------------ data
@@ -114,46 +115,83 @@ class ALSTrainer:
 
     # ------------------------------------------------------------ training
 
-    def _half_step(self, csr, fixed_local: torch.Tensor,
-                   n_fixed: int) -> torch.Tensor:
-        """One half-iteration: all-gather the fixed side, fused solve.
+    def _tsync(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        return time.time()
 
-        Set PIO_PHASE_TIMES=1 to accumulate per-phase wall times
-        (gather / solve) into self.phase_times — the diagnostic for
-        reading multi-GPU scaling results."""
-        timing = os.environ.get("PIO_PHASE_TIMES") == "1"
-        if timing:
-            if self.device.type == "cuda":
-                torch.cuda.synchronize()
-            t0 = time.time()
+    def _solve_half(self, csr, fixed_full: torch.Tensor,
+                    yty, n_local: int, n_out_total: int):
+        """Chunked fused solve into a zero-padded [max_rows, f] block;
+        each chunk's output all-gather is issued asynchronously right
+        after the chunk's kernels, so the collective for the NEXT
+        half-step hides under the remaining chunks' solve (VERDICT r1
+        item 2 — MLlib's shuffle/compute interleave).
+        Returns (local_rows_view, ChunkedGather with pushes issued)."""
         wire = (torch.bfloat16 if self.p.gather_dtype == "bf16" else None)
-        fixed_full = pdist.all_gather_rows(fixed_local, n_fixed,
-                                           wire_dtype=wire)
-        if timing:
-            if self.device.type == "cuda":
-                torch.cuda.synchronize()
-            t1 = time.time()
-        # YtY from the LOCAL shard + an FxF all-reduce: the post-gather
-        # Gramian would redo a world-size-times-larger GEMM on every rank
-        yty = (pdist.all_reduce_sum(als_ops.gramian(fixed_local))
-               if self.p.implicit else None)
+        g = pdist.ChunkedGather(n_out_total, self.p.rank, self.device,
+                                wire_dtype=wire)
+        n_chunks = (int(os.environ.get("PIO_GATHER_CHUNKS", "4"))
+                    if pdist.get_world_size() > 1 else 1)
         indptr, indices, values = csr
-        out = als_ops.als_solve(
-            indptr, indices, values, fixed_full, YtY=yty,
-            lam=self.p.lambda_, alpha=self.p.alpha,
-            implicit=self.p.implicit, wr_scale=not self.p.implicit)
-        if timing:
-            if self.device.type == "cuda":
-                torch.cuda.synchronize()
-            t2 = time.time()
-            self.phase_times["gather_s"] += t1 - t0
-            self.phase_times["solve_s"] += t2 - t1
-        return out
+        lv = None
+        if self.p.implicit and fixed_full.is_cuda:
+            lv = als_ops.prepare_lv(fixed_full, yty, self.p.lambda_)
+        step_rows = (g.max_rows + n_chunks - 1) // n_chunks
+        for c in range(n_chunks):
+            c0 = c * step_rows
+            c1 = min(g.max_rows, c0 + step_rows)
+            if c0 >= c1:
+                break
+            rl, rh = min(c0, n_local), min(c1, n_local)
+            if rh > rl:
+                als_ops.als_solve(
+                    indptr, indices, values, fixed_full, YtY=yty,
+                    lam=self.p.lambda_, alpha=self.p.alpha,
+                    implicit=self.p.implicit,
+                    wr_scale=not self.p.implicit, lv=lv,
+                    row_range=(rl, rh), out=g.src[rl:rh])
+            g.push(c0, c1)
+        return g.src[:n_local], g
 
     def step(self) -> None:
-        """One full ALS iteration (user half-step then item half-step)."""
-        self.X = self._half_step(self.user_csr, self.Y, self.n_items)
-        self.Y = self._half_step(self.item_csr, self.X, self.n_users)
+        """One full ALS iteration (user half-step then item half-step).
+
+        Set PIO_PHASE_TIMES=1 to accumulate per-phase wall times into
+        self.phase_times: 'solve_s' covers the chunked solves (with the
+        overlapped gather issuance inside), 'gather_s' is the EXPOSED
+        communication wait (pending-gather finishes + YtY reduces)."""
+        timing = os.environ.get("PIO_PHASE_TIMES") == "1"
+        implicit = self.p.implicit
+        t0 = self._tsync() if timing else 0.0
+        # full Y for the user half: consume the gather pipelined from the
+        # previous iteration's item half (first iteration gathers fresh)
+        if self._pending_y is not None:
+            Yfull = self._pending_y.finish()
+            self._pending_y = None
+        else:
+            wire = (torch.bfloat16 if self.p.gather_dtype == "bf16"
+                    else None)
+            Yfull = pdist.all_gather_rows(self.Y, self.n_items,
+                                          wire_dtype=wire)
+        # YtY from the LOCAL shard + an FxF all-reduce: the post-gather
+        # Gramian would redo a world-size-times-larger GEMM on every rank
+        yty = (pdist.all_reduce_sum(als_ops.gramian(self.Y))
+               if implicit else None)
+        t1 = self._tsync() if timing else 0.0
+        self.X, gx = self._solve_half(
+            self.user_csr, Yfull, yty, self.u_hi - self.u_lo, self.n_users)
+        t2 = self._tsync() if timing else 0.0
+        Xfull = gx.finish()
+        yty2 = (pdist.all_reduce_sum(als_ops.gramian(self.X))
+                if implicit else None)
+        t3 = self._tsync() if timing else 0.0
+        self.Y, self._pending_y = self._solve_half(
+            self.item_csr, Xfull, yty2, self.i_hi - self.i_lo, self.n_items)
+        if timing:
+            t4 = self._tsync()
+            self.phase_times["gather_s"] += (t1 - t0) + (t3 - t2)
+            self.phase_times["solve_s"] += (t2 - t1) + (t4 - t3)
 
     def fit(self) -> Tuple[torch.Tensor, torch.Tensor]:
         """Train and return the FULL (n_users x f, n_items x f) factor
@@ -172,10 +210,17 @@ class ALSTrainer:
     # ------------------------------------------------------------ model out
 
     def gather_factors(self) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Full (X, Y) on every rank (checkpoint/serving path)."""
+        """Full (X, Y) on every rank (checkpoint/serving path). Consumes
+        the pipelined Y gather from the last step when one is pending
+        (its wire-dtype rounding, if enabled, applies — same semantics
+        as the factors every rank trains against)."""
         X = pdist.all_gather_rows(self.X, self.n_users)
-        Y = pdist.all_gather_rows(self.Y, self.n_items)
-        return X, Y
+        if self._pending_y is not None:
+            Y = self._pending_y.finish()
+            self._pending_y = None
+        else:
+            Y = pdist.all_gather_rows(self.Y, self.n_items)
+        return X.contiguous(), Y.contiguous()
 
     def local_nnz(self) -> int:
         return int(self.user_csr[1].numel() + self.item_csr[1].numel())

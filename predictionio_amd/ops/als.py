@@ -49,7 +49,10 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
               values: torch.Tensor, Y: torch.Tensor,
               YtY: Optional[torch.Tensor] = None,
               lam: float = 0.01, alpha: float = 1.0,
-              implicit: bool = False, wr_scale: bool = True) -> torch.Tensor:
+              implicit: bool = False, wr_scale: bool = True,
+              lv: Optional[tuple] = None,
+              row_range: Optional[Tuple[int, int]] = None,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Solve all rows of one ALS half-iteration.
 
     explicit: (sum y y^T + lam*nnz*I) x = sum r*y      (ALS-WR, like MLlib)
@@ -61,11 +64,20 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
     three phases: the Woodbury kernel emits whitened Z rows, one rocBLAS
     triangular solve maps Z -> X for all rows, then the dense kernel fills
     the big (nnz > 32) rows directly.
+
+    lv: optional precomputed (Linv, V) from prepare_lv() — hoist it when
+    solving in row chunks (the multi-GPU gather/solve overlap path).
+    row_range: solve only rows [lo, hi) of the CSR (indices/values stay
+    whole; indptr offsets are absolute). out: optional [rows, f] target
+    written in place (enables chunked solves into one padded buffer).
     """
     if implicit:
         wr_scale = False  # Hu-Koren regularizes with plain lambda*I
         if YtY is None:
             YtY = gramian(Y)
+    if row_range is not None:
+        lo, hi = row_range
+        indptr = indptr[lo:hi + 1]
     if Y.is_cuda:
         from predictionio_amd.ops import hip_ext
         ext = hip_ext()
@@ -89,22 +101,53 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
                 raise ValueError(
                     f"CSR column id out of range: [{int(ix.min())}, {mx}]"
                     f" vs {Y.shape[0]} factor rows")
+        direct_out = (out if out is not None and pf == f
+                      and out.is_contiguous() else None)
         if implicit and pf <= 128:
-            Linv, V = woodbury_lv(Yp, YtYp, lam)
+            if lv is not None:
+                Linv, V = lv
+            else:
+                Linv, V = woodbury_lv(Yp, YtYp, lam)
             Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
                               float(alpha), True, False, 1, None)
             # X = Z L^-1 for all rows (Woodbury rows hold z; big rows get
             # overwritten by the dense pass next)
-            X = (Z @ Linv).contiguous()
+            if direct_out is not None:
+                X = direct_out
+                torch.matmul(Z, Linv, out=X)
+            else:
+                X = (Z @ Linv).contiguous()
             ext.als_solve(ip, ix, vv, Yp, YtYp, None, float(lam),
                           float(alpha), True, False, 2, X)
         else:
             X = ext.als_solve(ip, ix, vv, Yp, YtYp, None, float(lam),
                               float(alpha), bool(implicit), bool(wr_scale),
-                              0, None)
-        return X[:, :f].contiguous() if pf != f else X
-    return als_solve_ref(indptr, indices, values, Y, YtY, lam, alpha,
-                         implicit, wr_scale)
+                              0, direct_out)
+        res = X[:, :f].contiguous() if pf != f else X
+        if out is not None and res.data_ptr() != out.data_ptr():
+            out.copy_(res)
+            return out
+        return res
+    res = als_solve_ref(indptr, indices, values, Y, YtY, lam, alpha,
+                        implicit, wr_scale)
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
+
+
+def prepare_lv(Y: torch.Tensor, YtY: torch.Tensor, lam: float):
+    """Precompute the (Linv, V) Woodbury pair once per half-iteration so
+    chunked solves (gather/solve overlap) don't redo the whitening GEMM
+    per chunk. Pads to the supported rank like als_solve."""
+    f = Y.shape[1]
+    pf = pad_rank(f)
+    Yp = (Y if pf == f
+          else torch.nn.functional.pad(Y, (0, pf - f))).contiguous()
+    YtYp = (YtY if pf == f
+            else torch.nn.functional.pad(YtY, (0, pf - f, 0, pf - f))
+            ).contiguous()
+    return woodbury_lv(Yp, YtYp, lam)
 
 
 def als_solve_ref(indptr, indices, values, Y, YtY=None, lam=0.01, alpha=1.0,
