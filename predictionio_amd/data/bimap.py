@@ -84,6 +84,17 @@ class BiMap:
 
     string_long = string_int
 
+    @classmethod
+    def from_uniques(cls, uniques) -> "BiMap":
+        """BiMap over an already-distinct key sequence (e.g.
+        pandas.factorize uniques) — skips the uniqueness re-check."""
+        if isinstance(uniques, np.ndarray):
+            uniques = uniques.tolist()
+        b = cls.__new__(cls)
+        b._fwd = dict(zip(uniques, range(len(uniques))))
+        b._inv = None
+        return b
+
     def map_array(self, keys: List) -> np.ndarray:
         """Vectorized lookup → int64 array; raises KeyError on miss."""
         return np.fromiter((self._fwd[k] for k in keys), dtype=np.int64,

@@ -71,6 +71,25 @@ def find(app_name: str,
         target_entity_id=target_entity_id, limit=limit, reversed=reversed))
 
 
+def find_columns(app_name: str,
+                 channel_name: Optional[str] = None,
+                 start_time: Optional[datetime] = None,
+                 until_time: Optional[datetime] = None,
+                 entity_type: Optional[str] = None,
+                 event_names: Optional[List[str]] = None,
+                 target_entity_type: Any = UNSET,
+                 property_fields=()) -> Dict[str, list]:
+    """Training-time bulk COLUMNAR read (PEventStore.find semantics with
+    columnar output — the event-store→device ingest path; see
+    LEvents.find_columns)."""
+    app_id, channel_id = app_name_to_id(app_name, channel_name)
+    return storage.get_p_events().find_columns(
+        app_id=app_id, channel_id=channel_id, start_time=start_time,
+        until_time=until_time, entity_type=entity_type,
+        event_names=event_names, target_entity_type=target_entity_type,
+        property_fields=property_fields)
+
+
 def aggregate_properties(app_name: str, entity_type: str,
                          channel_name: Optional[str] = None,
                          start_time: Optional[datetime] = None,

@@ -175,6 +175,22 @@ class EngineInstances:
                              engine_variant: str) -> Optional[EngineInstance]:
         raise NotImplementedError
 
+    def get_latest_completed_by_factory(
+            self, engine_factory: str,
+            engine_variant: Optional[str] = None
+    ) -> Optional["EngineInstance"]:
+        """Deploy-time resolution (commands/Engine.deploy :208-245 /
+        EngineInstances.getLatestCompleted EngineInstances.scala:69) —
+        in the DAO so backends can push the filter into the store
+        instead of the server scanning get_all(). Default: scan."""
+        cands = [i for i in self.get_all()
+                 if i.status == "COMPLETED"
+                 and i.engine_factory == engine_factory
+                 and (engine_variant is None
+                      or i.engine_variant == engine_variant)]
+        cands.sort(key=lambda i: i.start_time)
+        return cands[-1] if cands else None
+
     def get_completed(self, engine_id: str, engine_version: str,
                       engine_variant: str) -> List[EngineInstance]:
         raise NotImplementedError
@@ -264,6 +280,41 @@ class LEvents:
         a string = must equal. limit None = all, -1 = all. reversed sorts
         by eventTime descending."""
         raise NotImplementedError
+
+    def find_columns(self, app_id: int, channel_id: Optional[int] = None,
+                     start_time: Optional[datetime] = None,
+                     until_time: Optional[datetime] = None,
+                     entity_type: Optional[str] = None,
+                     event_names: Optional[List[str]] = None,
+                     target_entity_type: Any = UNSET,
+                     property_fields: Sequence[str] = ()
+                     ) -> Dict[str, list]:
+        """Bulk columnar read for training ingest — PEvents.find
+        semantics (PEvents.scala:80-89) returning parallel COLUMNS
+        instead of per-event objects, so 10^7-10^9-event reads skip
+        Python object construction entirely. Returns
+        {'event': [str], 'entity_id': [str], 'target_entity_id':
+        [str|None], 'event_time_ms': [int]} plus one column per
+        requested property field (value or None). Rows are
+        eventTime-ascending (the dedup contracts — latest-wins — rely
+        on it). Backends override with store-side columnar scans; this
+        default walks find()."""
+        out: Dict[str, list] = {"event": [], "entity_id": [],
+                                "target_entity_id": [], "event_time_ms": []}
+        for f in property_fields:
+            out[f] = []
+        for e in self.find(app_id=app_id, channel_id=channel_id,
+                           start_time=start_time, until_time=until_time,
+                           entity_type=entity_type, event_names=event_names,
+                           target_entity_type=target_entity_type):
+            out["event"].append(e.event)
+            out["entity_id"].append(e.entity_id)
+            out["target_entity_id"].append(e.target_entity_id)
+            out["event_time_ms"].append(
+                int(e.event_time.timestamp() * 1000))
+            for f in property_fields:
+                out[f].append(e.properties.get_opt(f))
+        return out
 
     def aggregate_properties(self, app_id: int, entity_type: str,
                              channel_id: Optional[int] = None,

@@ -356,6 +356,18 @@ class SQLiteEngineInstances(base.EngineInstances):
         done = self.get_completed(engine_id, engine_version, engine_variant)
         return done[0] if done else None
 
+    def get_latest_completed_by_factory(self, engine_factory,
+                                        engine_variant=None):
+        q = (f"SELECT {self.COLS} FROM {self.TABLE} WHERE status='COMPLETED'"
+             " AND engineFactory=?")
+        args = [engine_factory]
+        if engine_variant is not None:
+            q += " AND engineVariant=?"
+            args.append(engine_variant)
+        q += " ORDER BY startTime DESC LIMIT 1"
+        row = self.c.conn().execute(q, args).fetchone()
+        return _row_to_ei(row) if row else None
+
     def update(self, i: EngineInstance) -> bool:
         conn = self.c.conn()
         row = _ei_to_row(i)
@@ -574,6 +586,59 @@ class SQLiteLEvents(base.LEvents):
             return False
         conn.commit()
         return cur.rowcount > 0
+
+    def find_columns(self, app_id: int, channel_id: Optional[int] = None,
+                     start_time: Optional[datetime] = None,
+                     until_time: Optional[datetime] = None,
+                     entity_type: Optional[str] = None,
+                     event_names: Optional[List[str]] = None,
+                     target_entity_type: Any = UNSET,
+                     property_fields: Sequence[str] = ()):
+        """Store-side columnar scan: one SELECT with json_extract()
+        pulling the requested property fields at C speed — no per-event
+        Python objects (the bulk-ingest path of SURVEY §2.9 K5; the
+        reference's analog streams rows into executors,
+        PEvents.scala:80-89)."""
+        t = self._table(app_id, channel_id)
+        clauses, args = [], []
+        if start_time is not None:
+            clauses.append("eventTime >= ?")
+            args.append(_dt_to_ms(start_time))
+        if until_time is not None:
+            clauses.append("eventTime < ?")
+            args.append(_dt_to_ms(until_time))
+        if entity_type is not None:
+            clauses.append("entityType = ?")
+            args.append(entity_type)
+        if event_names:
+            clauses.append(
+                "event IN (" + ",".join("?" * len(event_names)) + ")")
+            args.extend(event_names)
+        if target_entity_type is not UNSET:
+            if target_entity_type is None:
+                clauses.append("targetEntityType IS NULL")
+            else:
+                clauses.append("targetEntityType = ?")
+                args.append(target_entity_type)
+        sel = "event, entityId, targetEntityId, eventTime"
+        for f in property_fields:
+            if not f.replace("_", "").isalnum():
+                raise ValueError(f"bad property field name: {f}")
+            sel += f", json_extract(properties, '$.{f}')"
+        q = f"SELECT {sel} FROM {t}"
+        if clauses:
+            q += " WHERE " + " AND ".join(clauses)
+        q += " ORDER BY eventTime ASC"
+        try:
+            rows = self.c.conn().execute(q, args).fetchall()
+        except sqlite3.OperationalError:
+            rows = []
+        names = (["event", "entity_id", "target_entity_id",
+                  "event_time_ms"] + list(property_fields))
+        if not rows:
+            return {n: [] for n in names}
+        cols = list(zip(*rows))  # C-speed transpose
+        return {n: list(c) for n, c in zip(names, cols)}
 
     def find(self, app_id: int, channel_id: Optional[int] = None,
              start_time: Optional[datetime] = None,

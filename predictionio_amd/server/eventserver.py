@@ -263,15 +263,43 @@ def create_app(stats_on: bool = False,
                 {"message": "Batch request must have less than or equal to "
                             f"{MAX_BATCH_SIZE} events"},
                 status_code=400)
+        # validate/authorize every item first, then ONE bulk DB write for
+        # the valid ones (the reference batches via async futures,
+        # LEvents.futureInsertBatch LEvents.scala:106-112; here the DAO's
+        # insert_batch is a single executemany). Per-event statuses keep
+        # their order.
         results = []
-        for item in items:
+        valid: list = []  # (result_slot_index, Event)
+        for i, item in enumerate(items):
             try:
-                status, body = _insert_one(item, auth)
+                event = Event.from_json(item)
+                validate_event(event)
             except Exception as e:
-                status, body = 500, {"message": str(e)}
-            results.append({"status": status, **body})
-            if stats_on:
-                stats.bookkeeping(auth.app_id, status)
+                results.append({"status": 400, "message": str(e)})
+                continue
+            if auth.events and event.event not in auth.events:
+                results.append({"status": 403,
+                                "message": f"{event.event} events are not "
+                                           "allowed"})
+                continue
+            blocked = _notify(event, auth)
+            if blocked is not None:
+                results.append({"status": 403, "message": blocked})
+                continue
+            results.append(None)
+            valid.append((len(results) - 1, event))
+        if valid:
+            try:
+                ids = levents.insert_batch([e for _, e in valid],
+                                           auth.app_id, auth.channel_id)
+                for (slot, _), eid in zip(valid, ids):
+                    results[slot] = {"status": 201, "eventId": eid}
+            except Exception as e:
+                for slot, _ in valid:
+                    results[slot] = {"status": 500, "message": str(e)}
+        if stats_on:
+            for r in results:
+                stats.bookkeeping(auth.app_id, r["status"])
         return results
 
     @app.get("/stats.json")

@@ -101,14 +101,12 @@ def _load_state(config: ServerConfig):
         inst = instances.get(config.engine_instance_id)
     else:
         # latest COMPLETED instance of this engine factory (+variant if
-        # one was given) — commands/Engine.deploy resolution (:208-245)
-        cands = [i for i in instances.get_all()
-                 if i.status == "COMPLETED"
-                 and i.engine_factory == config.engine_factory
-                 and (config.engine_variant in ("", "engine.json")
-                      or i.engine_variant == config.engine_variant)]
-        cands.sort(key=lambda i: i.start_time)
-        inst = cands[-1] if cands else None
+        # one was given) — resolved in the DAO like the reference
+        # (EngineInstances.getLatestCompleted, EngineInstances.scala:69)
+        variant = (None if config.engine_variant in ("", "engine.json")
+                   else config.engine_variant)
+        inst = instances.get_latest_completed_by_factory(
+            config.engine_factory, variant)
     if inst is None:
         raise RuntimeError(
             f"No COMPLETED engine instance found for "

@@ -267,3 +267,53 @@ class TestConcurrentAccess:
         le = mem_storage.get_l_events()
         le.init(app_id)
         self._hammer(le, app_id)
+
+
+class TestFindColumns:
+    """Columnar bulk read (LEvents.find_columns): the sqlite store-side
+    scan must agree with the base-class default (walking find())."""
+
+    @pytest.fixture()
+    def events(self, mem_storage):
+        le = mem_storage.get_l_events()
+        le.init(1)
+        le.insert_batch([
+            mk("rate", "u1", 0, target_entity_type="item",
+               target_entity_id="i1",
+               properties=DataMap({"rating": 4.5})),
+            mk("buy", "u1", 1, target_entity_type="item",
+               target_entity_id="i2"),
+            mk("rate", "u2", 2, target_entity_type="item",
+               target_entity_id="i1",
+               properties=DataMap({"rating": 2.0})),
+            mk("view", "u3", 3),
+            mk("$set", "i1", 4, etype="item"),
+        ], 1)
+        return le
+
+    def test_columns_match_find(self, events):
+        from predictionio_amd.data.storage import base as sbase
+        cols = events.find_columns(
+            app_id=1, entity_type="user", event_names=["rate", "buy"],
+            target_entity_type="item", property_fields=["rating"])
+        ref = sbase.LEvents.find_columns(
+            events, app_id=1, entity_type="user",
+            event_names=["rate", "buy"], target_entity_type="item",
+            property_fields=["rating"])
+        assert cols["event"] == ref["event"] == ["rate", "buy", "rate"]
+        assert cols["entity_id"] == ref["entity_id"] == ["u1", "u1", "u2"]
+        assert (cols["target_entity_id"] == ref["target_entity_id"]
+                == ["i1", "i2", "i1"])
+        assert cols["rating"] == ref["rating"] == [4.5, None, 2.0]
+        assert cols["event_time_ms"] == ref["event_time_ms"]
+
+    def test_all_events_no_filters(self, events):
+        cols = events.find_columns(app_id=1)
+        assert len(cols["event"]) == 5
+        # ascending event time
+        assert cols["event_time_ms"] == sorted(cols["event_time_ms"])
+
+    def test_bad_property_field_rejected(self, events):
+        with pytest.raises(ValueError):
+            events.find_columns(app_id=1,
+                                property_fields=["x'); DROP TABLE t;--"])
