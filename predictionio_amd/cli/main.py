@@ -253,6 +253,17 @@ def adminserver(ip, port):
     run(host=ip, port=port)
 
 
+@cli.command()
+@click.option("--ip", default="0.0.0.0")
+@click.option("--port", default=7072)
+def storageserver(ip, port):
+    """Storage server daemon — serves this process's configured storage
+    (sqlite WAL by default) to remote-backend clients; the framework's
+    client-server database tier (the reference's JDBC PostgreSQL role)."""
+    from predictionio_amd.server.storageserver import run
+    run(host=ip, port=port)
+
+
 # ----------------------------------------------------------------- app
 
 @cli.group()

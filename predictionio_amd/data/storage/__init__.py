@@ -97,6 +97,11 @@ def _get_client(source_name: str, cfg: dict):
             )
             client = FsspecClient(cfg.get(
                 "path", os.path.join(_base_dir(), "models")))
+        elif typ == "remote":
+            from predictionio_amd.data.storage.remote import RemoteClient
+            client = RemoteClient(
+                cfg.get("url", "http://127.0.0.1:7072"),
+                timeout=float(cfg.get("timeout", "30")))
         else:
             raise StorageError(f"Unknown storage source type: {typ}")
         _sources[key] = client
@@ -133,6 +138,15 @@ def _dao(repo: str, kind: str):
     elif typ == "fsspec":
         from predictionio_amd.data.storage import fsspec_store as be
         table = {"models": be.FsspecModels}
+    elif typ == "remote":
+        from predictionio_amd.data.storage import remote as be
+        table = {
+            "apps": be.RemoteApps, "accesskeys": be.RemoteAccessKeys,
+            "channels": be.RemoteChannels,
+            "engineinstances": be.RemoteEngineInstances,
+            "evaluationinstances": be.RemoteEvaluationInstances,
+            "models": be.RemoteModels, "levents": be.RemoteLEvents,
+        }
     else:
         raise StorageError(f"Unknown storage type {typ}")
     if kind not in table:

@@ -531,12 +531,17 @@ class SQLiteLEvents(base.LEvents):
         self.c.close()
 
     def _row(self, e: Event):
-        eid = e.event_id or uuid.uuid4().hex
+        # hot ingest path: os.urandom-hex ids (uuid4() costs ~7 us in
+        # UUID-object overhead) and constant shortcuts for the empty
+        # properties/tags that dominate implicit-event streams
+        eid = e.event_id or os.urandom(16).hex()
+        props = e.properties.to_dict()
+        tags = e.tags
         return (eid, e.event, e.entity_type, e.entity_id,
                 e.target_entity_type, e.target_entity_id,
-                json.dumps(e.properties.to_dict()),
+                json.dumps(props) if props else "{}",
                 _dt_to_ms(e.event_time), str(e.event_time.tzinfo or "UTC"),
-                json.dumps(e.tags), e.pr_id,
+                json.dumps(tags) if tags else "[]", e.pr_id,
                 _dt_to_ms(e.creation_time), str(e.creation_time.tzinfo or "UTC"))
 
     def insert(self, event: Event, app_id: int,
