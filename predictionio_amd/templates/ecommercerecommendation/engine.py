@@ -142,7 +142,8 @@ class ECommModel:
     def __init__(self, rank: int, user_features: torch.Tensor,
                  item_factors: torch.Tensor, item_factors_norm: torch.Tensor,
                  user_map: BiMap, item_map: BiMap,
-                 items: Dict[str, Item], popular_count: Dict[int, int]):
+                 items: Dict[str, Item], popular_count: Dict[int, int],
+                 category_masks=None):
         self.rank = rank
         self.user_features = user_features
         self.item_factors = item_factors
@@ -152,6 +153,16 @@ class ECommModel:
         self.items = items
         self.popular_count = popular_count
         self.item_inv = item_map.inverse_array()
+        self.category_masks = category_masks
+
+    def cat_masks(self):
+        """Device category masks (train-time; lazily rebuilt for models
+        persisted before round 2)."""
+        if self.category_masks is None:
+            from predictionio_amd.templates.common import CategoryMasks
+            self.category_masks = CategoryMasks.build(
+                self.items, self.item_map, self.item_factors.device)
+        return self.category_masks
 
 
 class ECommAlgorithm(Algorithm):
@@ -196,8 +207,10 @@ class ECommAlgorithm(Algorithm):
         for e in pd.buy_events:
             pop[item_map[e.item]] += 1
         Yn = torch.nn.functional.normalize(Y, dim=1, eps=1e-9)
+        from predictionio_amd.templates.common import CategoryMasks
+        cm = CategoryMasks.build(pd.items, item_map, Y.device)
         return ECommModel(p.rank, X, Y, Yn, user_map, item_map, pd.items,
-                          dict(pop))
+                          dict(pop), cm)
 
     # ------------------------------------------------------------ filters
 
@@ -228,28 +241,20 @@ class ECommAlgorithm(Algorithm):
 
     def _mask(self, model: ECommModel, q: Query, black: Set[str]
               ) -> torch.Tensor:
+        """Device-resident filters: category membership from train-time
+        masks, white/black lists as small index_put ops — query cost is
+        flat in catalog size."""
+        from predictionio_amd.templates.common import ids_tensor
+        dev = model.item_factors.device
         n = len(model.item_map)
-        mask = torch.zeros(n, dtype=torch.uint8)
+        mask = torch.zeros(n, dtype=torch.uint8, device=dev)
         if q.white_list is not None:
             mask[:] = 1
-            for it in q.white_list:
-                i = model.item_map.get(it)
-                if i is not None:
-                    mask[i] = 0
+            mask[ids_tensor(q.white_list, model.item_map, dev)] = 0
         if q.categories is not None:
-            cats = set(q.categories)
-            cm = torch.ones(n, dtype=torch.uint8)
-            for it, meta in model.items.items():
-                if meta.categories and cats & set(meta.categories):
-                    i = model.item_map.get(it)
-                    if i is not None:
-                        cm[i] = 0
-            mask |= cm
-        for it in black:
-            i = model.item_map.get(it)
-            if i is not None:
-                mask[i] = 1
-        return mask.to(model.item_factors.device)
+            mask |= model.cat_masks().banned_outside(q.categories).to(dev)
+        mask[ids_tensor(black, model.item_map, dev)] = 1
+        return mask
 
     def _weight_groups(self) -> List[WeightGroup]:
         """adjust-score variant: [{"items": [...], "weight": w}]

@@ -132,11 +132,21 @@ class SimilarModel:
     """Normalized item factors + maps + categories (P2L local model)."""
 
     def __init__(self, item_factors_norm: torch.Tensor, item_map: BiMap,
-                 items: Dict[str, Item]):
+                 items: Dict[str, Item], category_masks=None):
         self.item_factors_norm = item_factors_norm  # L2-normalized rows
         self.item_map = item_map
         self.items = items
         self.item_inv = item_map.inverse_array()
+        self.category_masks = category_masks
+
+    def cat_masks(self):
+        """Device category masks (built at train time; lazily rebuilt
+        for models persisted before round 2)."""
+        if self.category_masks is None:
+            from predictionio_amd.templates.common import CategoryMasks
+            self.category_masks = CategoryMasks.build(
+                self.items, self.item_map, self.item_factors_norm.device)
+        return self.category_masks
 
 
 class ALSAlgorithm(Algorithm):
@@ -171,43 +181,34 @@ class ALSAlgorithm(Algorithm):
         _, Y = train_als(users, items, vals, len(user_map), len(item_map),
                          p, device=device)
         Yn = torch.nn.functional.normalize(Y, dim=1, eps=1e-9)
-        return SimilarModel(Yn, item_map, pd.items)
+        from predictionio_amd.templates.common import CategoryMasks
+        cm = CategoryMasks.build(pd.items, item_map, Yn.device)
+        return SimilarModel(Yn, item_map, pd.items, cm)
 
-    def _masks(self, model: SimilarModel, q: Query, dev) -> Optional[torch.Tensor]:
+    def _masks(self, model: SimilarModel, q: Query,
+               dev) -> Optional[torch.Tensor]:
         """uint8 banned mask from white/black/category filters
-        (ALSAlgorithm.scala:245-271 isCandidateItem)."""
+        (ALSAlgorithm.scala:245-271 isCandidateItem). Device-resident:
+        category membership comes from train-time masks, white/black
+        lists are small index_put ops — nothing scans the catalog."""
+        from predictionio_amd.templates.common import ids_tensor
         n = len(model.item_map)
         mask = None
         if q.white_list is not None:
-            mask = torch.ones(n, dtype=torch.uint8)
-            for it in q.white_list:
-                i = model.item_map.get(it)
-                if i is not None:
-                    mask[i] = 0
+            mask = torch.ones(n, dtype=torch.uint8, device=dev)
+            mask[ids_tensor(q.white_list, model.item_map, dev)] = 0
         if q.categories is not None:
-            cm = torch.ones(n, dtype=torch.uint8)
-            cats = set(q.categories)
-            for it, meta in model.items.items():
-                if meta.categories and cats & set(meta.categories):
-                    i = model.item_map.get(it)
-                    if i is not None:
-                        cm[i] = 0
+            cm = model.cat_masks().banned_outside(q.categories).to(dev)
             mask = cm if mask is None else (mask | cm)
         if q.black_list:
             if mask is None:
-                mask = torch.zeros(n, dtype=torch.uint8)
-            for it in q.black_list:
-                i = model.item_map.get(it)
-                if i is not None:
-                    mask[i] = 1
+                mask = torch.zeros(n, dtype=torch.uint8, device=dev)
+            mask[ids_tensor(q.black_list, model.item_map, dev)] = 1
         # query items themselves are never returned (reference excludes them)
         if mask is None:
-            mask = torch.zeros(n, dtype=torch.uint8)
-        for it in q.items:
-            i = model.item_map.get(it)
-            if i is not None:
-                mask[i] = 1
-        return mask.to(dev)
+            mask = torch.zeros(n, dtype=torch.uint8, device=dev)
+        mask[ids_tensor(q.items, model.item_map, dev)] = 1
+        return mask
 
     def predict(self, model: SimilarModel, query) -> PredictedResult:
         q = query if isinstance(query, Query) else Query.from_json(query)
