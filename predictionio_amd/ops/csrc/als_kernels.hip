@@ -735,6 +735,226 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Dual-row Woodbury solver (round 2): TWO rows per wave, one per 32-lane
+// half. The phase probe showed the single-row kernel issue-bound with the
+// register Cholesky M-solve at 66% of time while only n <= 32 of 64 lanes
+// were active — packing a second row into the idle half halves the
+// per-row issue count of the solve (both halves factor concurrently in
+// one instruction stream; __shfl with a computed (lane & 32) + k source
+// stays within each half). Staging/G/emit keep full-wave cooperation per
+// row, so their per-row cost is unchanged.
+// ---------------------------------------------------------------------------
+
+template <int F, int NW, int NLO>
+__global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
+    const long long* __restrict__ indptr,
+    const int* __restrict__ indices,
+    const float* __restrict__ values,
+    const float* __restrict__ Y,
+    const float* __restrict__ V,
+    float* __restrict__ X,
+    int n_rows,
+    float lambda,
+    float alpha,
+    int implicit_mode,
+    int wr_scale,
+    unsigned long long* prof)
+{
+  constexpr int FP = F + 4;
+  constexpr int MP = NW + 4;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int h = lane >> 5;    // wave half = which of the two rows
+  const int r = lane & 31;    // sub-lane within the half
+
+  __shared__ float Yl[2][2][NW][FP];   // [wave][half]
+  __shared__ float M[2][2][NW][MP];
+  __shared__ float tv[2][2][NW];
+  __shared__ float dv[2][2][NW];
+
+  const float* src = implicit_mode ? V : Y;
+
+  for (long long base = ((long long)blockIdx.x * 2 + wave) * 2;
+       base < n_rows; base += (long long)gridDim.x * 4) {
+    // ---- per-half row bookkeeping (uniform across the wave) ----
+    int nh[2];
+    long long starth[2];
+    bool act[2];
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+      const long long row = base + hh;
+      if (row < n_rows) {
+        starth[hh] = indptr[row];
+        nh[hh] = (int)(indptr[row + 1] - starth[hh]);
+        act[hh] = nh[hh] <= NW && nh[hh] > NLO;
+        if (nh[hh] == 0 && NLO < 0) {
+          for (int e = lane; e < F; e += 64)
+            X[row * (long long)F + e] = 0.f;
+          act[hh] = false;
+        }
+      } else {
+        act[hh] = false;
+        nh[hh] = 0;
+        starth[hh] = 0;
+      }
+    }
+    if (!act[0] && !act[1]) continue;
+    const bool probe = prof != nullptr && (base & 1023) == 0 && lane == 0
+                       && act[0];
+    unsigned long long pt0 = 0, pt1 = 0, pt2 = 0, pt3 = 0;
+    if (probe) pt0 = wall_clock64();
+
+    // ---- stage both rows (full-wave per row, same as single-row) ----
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+      if (!act[hh]) continue;
+      const int n = nh[hh];
+      const long long start = starth[hh];
+      float* yl = &Yl[wave][hh][0][0];
+      if (lane < n)
+        tv[wave][hh][lane] = __int_as_float(indices[start + lane]);
+      wave_sync();
+      for (int c = 0; c < n; ++c) {
+        const long long col = __float_as_int(tv[wave][hh][c]);
+        for (int e = lane; e < F; e += 64)
+          yl[c * FP + e] = src[col * F + e];
+      }
+      if (lane < n) {
+        const float v = values[start + lane];
+        if (implicit_mode) {
+          const float ar = alpha * v;
+          const float d = sqrtf(ar > 1e-12f ? ar : 1e-12f);
+          dv[wave][hh][lane] = d;
+          tv[wave][hh][lane] = (1.f + ar) / d;
+        } else {
+          tv[wave][hh][lane] = v;
+        }
+      }
+    }
+    wave_sync();
+    if (probe) pt1 = wall_clock64();
+
+    // ---- G build: each half computes its own row's pair triangle ----
+    const int n = nh[h];           // this half's row size (per-lane)
+    {
+      const float* yl = &Yl[wave][h][0][0];
+      const float reg = wr_scale ? lambda * (float)n : lambda;
+      const int npairs = act[h] ? n * (n + 1) / 2 : 0;
+      int p = r, i = 0;
+      while (p >= n - i && i < n) { p -= n - i; ++i; }
+      int j = i + p;
+      for (int pp = r; pp < npairs; pp += 32) {
+        const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
+        const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
+        f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
+        float dot = acc4.x + acc4.y + acc4.z + acc4.w;
+        if (implicit_mode) {
+          dot *= dv[wave][h][i] * dv[wave][h][j];
+          if (i == j) dot += 1.f;
+        } else if (i == j) {
+          dot += reg;
+        }
+        M[wave][h][i][j] = dot;
+        M[wave][h][j][i] = dot;
+        j += 32;
+        while (j >= n && i < n) { ++i; j -= n - i; }
+      }
+    }
+    wave_sync();
+    if (probe) pt2 = wall_clock64();
+
+    // ---- Cholesky: both halves factor their M concurrently ----
+    float mr[NW];
+#pragma unroll
+    for (int q = 0; q < NW; ++q)
+      mr[q] = (r < n && q < n) ? M[wave][h][r][q] : 0.f;
+    float t = r < n ? tv[wave][h][r] : 0.f;
+    wave_sync();
+    const int hb = lane & 32;     // shfl base of this half
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {                // per-lane guard (halves may differ)
+        float mkk = __shfl(mr[k], hb + k);
+        mkk = mkk > 1e-30f ? mkk : 1e-30f;
+        const float dinv = rsqrtf(mkk);
+        const float ljk = r > k ? mr[k] * dinv
+                                : (r == k ? mkk * dinv : 0.f);
+        mr[k] = ljk;
+        if (r < NW) M[wave][h][k][r] = ljk;  // column k stored as row k
+      }
+      wave_sync();
+      if (k < n) {
+#pragma unroll
+        for (int j4 = (k + 1) & ~3; j4 < NW; j4 += 4) {
+          if (j4 < n) {
+            const float4 q4 =
+                *reinterpret_cast<const float4*>(&M[wave][h][k][j4]);
+            if (r > k) {
+              if (j4 + 0 > k) mr[j4 + 0] = fmaf(-mr[k], q4.x, mr[j4 + 0]);
+              if (j4 + 1 > k) mr[j4 + 1] = fmaf(-mr[k], q4.y, mr[j4 + 1]);
+              if (j4 + 2 > k) mr[j4 + 2] = fmaf(-mr[k], q4.z, mr[j4 + 2]);
+              if (j4 + 3 > k) mr[j4 + 3] = fmaf(-mr[k], q4.w, mr[j4 + 3]);
+            }
+          }
+        }
+      }
+      wave_sync();
+    }
+    // forward solve L z = rhs
+#pragma unroll
+    for (int k = 0; k < NW; ++k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], hb + k);
+        const float zk = __shfl(t, hb + k) / lkk;
+        if (r == k) t = zk;
+        else if (r > k) t = fmaf(-mr[k], zk, t);
+      }
+    }
+    // back solve L^T s = z
+    wave_sync();
+#pragma unroll
+    for (int k = NW - 1; k >= 0; --k) {
+      if (k < n) {
+        const float lkk = __shfl(mr[k], hb + k);
+        const float xk = __shfl(t, hb + k) / lkk;
+        if (r == k) t = xk;
+        else if (r < k) t = fmaf(-M[wave][h][r][k], xk, t);
+      }
+    }
+    if (implicit_mode) t *= dv[wave][h][r < n ? r : 0];
+    if (r < n && act[h]) tv[wave][h][r] = t;
+    wave_sync();
+    if (probe) pt3 = wall_clock64();
+
+    // ---- emit both rows (full-wave per row) ----
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+      if (!act[hh]) continue;
+      const float* yl = &Yl[wave][hh][0][0];
+      const int nn = nh[hh];
+      const long long row = base + hh;
+      for (int e = lane; e < F; e += 64) {
+        float x = 0.f;
+        for (int c = 0; c < nn; ++c)
+          x = fmaf(tv[wave][hh][c], yl[c * FP + e], x);
+        X[row * (long long)F + e] = x;
+      }
+    }
+    if (probe) {
+      const unsigned long long pt4 = wall_clock64();
+      atomicAdd(&prof[0], pt1 - pt0);
+      atomicAdd(&prof[1], pt2 - pt1);
+      atomicAdd(&prof[2], pt3 - pt2);
+      atomicAdd(&prof[3], pt4 - pt3);
+      atomicAdd(&prof[4], 1ull);
+    }
+    wave_sync();
+  }
+}
+
 // Variant flag for als_solve_kernel: skip rows the Woodbury kernel owns.
 
 // ---------------------------------------------------------------------------
@@ -773,8 +993,32 @@ extern "C" void launch_als_solve(
   hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,     \
                      stream, indptr, indices, values, Y, YtY, X, n_rows,     \
                      lambda, alpha, implicit_mode, wr_scale, skip)
+  // PIO_ALS_DUAL=1 routes the Woodbury rows through the dual-row kernel
+  // (two rows per wave) — A/B lever for the round-2 M-solve rework
+  static const bool use_dual = [] {
+    const char* e = getenv("PIO_ALS_DUAL");
+    return e != nullptr && e[0] == '1';
+  }();
+  long long wg4 = ((long long)n_rows + 3) / 4;
+  int grid_w2 = (int)(wg4 < (1 << 20) ? wg4 : (1 << 20));
 #define LAUNCH_WOODBURY(FF)                                                  \
-  if (woodbury) {                                                            \
+  if (woodbury && use_dual) {                                                \
+    /* dual for nnz<=24; the NW=32 dual instantiation spills (mr[32] x   */  \
+    /* dual-row bookkeeping exceeds the register budget) so rows 25-32   */  \
+    /* keep the single-row kernel                                        */  \
+    hipLaunchKernelGGL((als_woodbury2_kernel<FF, 20, -1>), dim3(grid_w2),    \
+                       dim3(128), 0, stream, indptr, indices, values, Y, V,  \
+                       X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
+                       prof);                                                \
+    hipLaunchKernelGGL((als_woodbury2_kernel<FF, 24, 20>), dim3(grid_w2),    \
+                       dim3(128), 0, stream, indptr, indices, values, Y, V,  \
+                       X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
+                       prof);                                                \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24>), dim3(grid_w),      \
+                       dim3(128), 0, stream, indptr, indices, values, Y, V,  \
+                       X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
+                       prof);                                                \
+  } else if (woodbury) {                                                     \
     hipLaunchKernelGGL((als_woodbury_kernel<FF, 20, -1>), dim3(grid_w),      \
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \
                        X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
