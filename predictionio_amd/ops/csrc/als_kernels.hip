@@ -456,8 +456,12 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
       const float ljk = lane > k ? acc[k] * dinv
                                  : (lane == k ? lkk * dinv : 0.f);
       // persist column k: Lc[k][j] = L[j][k]. Guard: for F < 64 lanes
-      // >= F would write past the row into the next column.
+      // >= F would write past the row into the next column. acc[k] is
+      // dead after this step — keep L[lane][k] in it so the forward
+      // solve runs from registers (round 2: the scalar Lc reads in the
+      // substitutions were the flagged ds_read_b32 round trips).
       if (lane < F) Lc[wave][k][lane] = ljk;
+      acc[k] = ljk;
       wave_sync();
       // trailing update reading the pivot column in b128 QUADS: the
       // scalar version compiled to ds_read_b32 + s_waitcnt lgkmcnt(0)
@@ -478,19 +482,33 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
       wave_sync();
     }
 
-    // ---- forward solve L z = b (z in b_reg); L read from LDS columns ----
+    // ---- forward solve L z = b — pure register/shfl chain: lane's row
+    //      L[lane][k] lives in acc[k], the pivot L[k][k] is lane k's
+    //      acc[k] (no LDS round trips on the serial chain) ----
+#pragma unroll
     for (int k = 0; k < F; ++k) {
-      const float zk = __shfl(b_reg, k) / Lc[wave][k][k];
+      const float zk = __shfl(b_reg, k) / __shfl(acc[k], k);
       if (lane == k) b_reg = zk;
-      else if (lane > k) b_reg = fmaf(-Lc[wave][k][lane], zk, b_reg);
+      else if (lane > k) b_reg = fmaf(-acc[k], zk, b_reg);
     }
-    // ---- back solve L^T x = z: lane j reads L[k][j] = Lc[j][k]
-    //      (Lc[c][r] stores L[r][c]; stride F+1 across lanes → no bank
-    //      conflicts) ----
-    for (int k = F - 1; k >= 0; --k) {
-      const float xk = __shfl(b_reg, k) / Lc[wave][k][k];
-      if (lane == k) b_reg = xk;
-      else if (lane < k) b_reg = fmaf(-Lc[wave][lane][k], xk, b_reg);
+    // ---- back solve L^T x = z: lane j needs L[k][j] = Lc[j][k], which
+    //      is CONSECUTIVE over k at fixed row j — read b128 quads so one
+    //      LDS round trip covers 4 serial steps ----
+    {
+      float4 lq = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int k = F - 1; k >= 0; --k) {
+        if ((k & 3) == 3 && lane < F)
+          lq = *reinterpret_cast<const float4*>(&Lc[wave][lane][k & ~3]);
+        const float xk = __shfl(b_reg, k) / __shfl(acc[k], k);
+        if (lane == k) b_reg = xk;
+        else if (lane < k) {
+          const float lv = (k & 3) == 0 ? lq.x
+                           : (k & 3) == 1 ? lq.y
+                           : (k & 3) == 2 ? lq.z : lq.w;
+          b_reg = fmaf(-lv, xk, b_reg);
+        }
+      }
     }
 
     if (lane < F) X[row * (long long)F + lane] = b_reg;
