@@ -58,10 +58,23 @@ class TrainingData(SanityCheck):
     items: Dict[str, Item]
     view_events: List[ItemEvent]
     buy_events: List[ItemEvent]
+    view_columns: Optional["EventColumns"] = None
+    buy_columns: Optional["EventColumns"] = None
 
     def sanity_check(self):
-        if not self.view_events:
+        if not self.view_events and not (
+                self.view_columns and len(self.view_columns)):
             raise ValueError("view events are empty")
+
+
+@dataclass
+class EventColumns:
+    """Columnar (user, item) event arrays from the bulk store read."""
+    users: Any   # np.ndarray object
+    items: Any   # np.ndarray object
+
+    def __len__(self):
+        return len(self.users)
 
 
 @dataclass
@@ -70,6 +83,8 @@ class PreparedData:
     items: Dict[str, Item]
     view_events: List[ItemEvent]
     buy_events: List[ItemEvent]
+    view_columns: Optional[EventColumns] = None
+    buy_columns: Optional[EventColumns] = None
 
 
 @dataclass
@@ -122,6 +137,22 @@ class DataSource(BaseDataSource):
                  for eid, pm in
                  event_store.aggregate_properties(app, "item").items()}
 
+        if self.params.get("columnar", True):
+            import numpy as np
+
+            def read_cols(name):
+                cols = event_store.find_columns(
+                    app, entity_type="user", event_names=[name],
+                    target_entity_type="item")
+                return EventColumns(
+                    users=np.asarray(cols["entity_id"], dtype=object),
+                    items=np.asarray(cols["target_entity_id"],
+                                     dtype=object))
+
+            return TrainingData(users, items, [], [],
+                                view_columns=read_cols("view"),
+                                buy_columns=read_cols("buy"))
+
         def read(name):
             return [ItemEvent(e.entity_id, e.target_entity_id,
                               e.event_time.timestamp())
@@ -135,7 +166,8 @@ class DataSource(BaseDataSource):
 class Preparator(BasePreparator):
     def prepare(self, td: TrainingData) -> PreparedData:
         return PreparedData(td.users, td.items, td.view_events,
-                            td.buy_events)
+                            td.buy_events, td.view_columns,
+                            td.buy_columns)
 
 
 class ECommModel:
@@ -170,26 +202,59 @@ class ECommAlgorithm(Algorithm):
     numIterations, lambda, alpha, [seed], [buyScore]."""
 
     def train(self, pd: PreparedData) -> ECommModel:
-        user_map = BiMap.string_int(
-            [e.user for e in pd.view_events]
-            + [e.user for e in pd.buy_events] + list(pd.users))
-        item_map = BiMap.string_int(
-            [e.item for e in pd.view_events]
-            + [e.item for e in pd.buy_events] + list(pd.items))
         # Reference parity: ECommAlgorithm.genMLlibRating builds MLlib
         # ratings from viewEvents ONLY (ECommAlgorithm.scala:168-205);
         # buys feed just the popularity counts. buyScore>0 blends buys
         # into training as an opt-in extension beyond the reference.
         buy_score = float(self.params.get("buyScore", 0.0))
-        evs = ([(e, 1.0) for e in pd.view_events]
-               + ([(e, buy_score) for e in pd.buy_events]
-                  if buy_score > 0.0 else []))
-        users = torch.tensor([user_map[e.user] for e, _ in evs],
-                             dtype=torch.int32)
-        items = torch.tensor([item_map[e.item] for e, _ in evs],
-                             dtype=torch.int32)
-        vals = torch.tensor([w for _, w in evs], dtype=torch.float32)
+        import numpy as np
         from predictionio_amd.ops import als as als_ops
+        if pd.view_columns is not None:
+            # columnar ingest: one joint factorize per id space gives the
+            # BiMaps AND the event codes in a single C pass
+            import pandas as pandas_
+            vc, bc = pd.view_columns, pd.buy_columns
+            nv, nb = len(vc), len(bc) if bc is not None else 0
+            ucat = np.concatenate(
+                [vc.users, bc.users if bc is not None else [],
+                 np.asarray(list(pd.users), dtype=object)])
+            icat = np.concatenate(
+                [vc.items, bc.items if bc is not None else [],
+                 np.asarray(list(pd.items), dtype=object)])
+            ucodes, uu = pandas_.factorize(ucat)
+            icodes, ui = pandas_.factorize(icat)
+            user_map = BiMap.from_uniques(uu)
+            item_map = BiMap.from_uniques(ui)
+            if buy_score > 0.0 and nb:
+                users_np = ucodes[:nv + nb]
+                items_np = icodes[:nv + nb]
+                vals_np = np.concatenate(
+                    [np.ones(nv, np.float32),
+                     np.full(nb, buy_score, np.float32)])
+            else:
+                users_np = ucodes[:nv]
+                items_np = icodes[:nv]
+                vals_np = np.ones(nv, np.float32)
+            users = torch.from_numpy(users_np.astype(np.int32))
+            items = torch.from_numpy(items_np.astype(np.int32))
+            vals = torch.from_numpy(vals_np)
+            buy_item_codes = icodes[nv:nv + nb]
+        else:
+            user_map = BiMap.string_int(
+                [e.user for e in pd.view_events]
+                + [e.user for e in pd.buy_events] + list(pd.users))
+            item_map = BiMap.string_int(
+                [e.item for e in pd.view_events]
+                + [e.item for e in pd.buy_events] + list(pd.items))
+            evs = ([(e, 1.0) for e in pd.view_events]
+                   + ([(e, buy_score) for e in pd.buy_events]
+                      if buy_score > 0.0 else []))
+            users = torch.tensor([user_map[e.user] for e, _ in evs],
+                                 dtype=torch.int32)
+            items = torch.tensor([item_map[e.item] for e, _ in evs],
+                                 dtype=torch.int32)
+            vals = torch.tensor([w for _, w in evs], dtype=torch.float32)
+            buy_item_codes = None
         users, items, vals = als_ops.aggregate_ratings(
             users, items, vals, len(item_map), "sum")
         p = ALSParams(
@@ -204,8 +269,12 @@ class ECommAlgorithm(Algorithm):
                          p, device=device)
         # popularity = buy counts (trainDefault :207-246)
         pop: Dict[int, int] = defaultdict(int)
-        for e in pd.buy_events:
-            pop[item_map[e.item]] += 1
+        if buy_item_codes is not None:
+            uniq, cnt = np.unique(buy_item_codes, return_counts=True)
+            pop = dict(zip(uniq.tolist(), cnt.tolist()))
+        else:
+            for e in pd.buy_events:
+                pop[item_map[e.item]] += 1
         Yn = torch.nn.functional.normalize(Y, dim=1, eps=1e-9)
         from predictionio_amd.templates.common import CategoryMasks
         cm = CategoryMasks.build(pd.items, item_map, Y.device)
