@@ -72,7 +72,7 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
-template <int F, bool PROF>
+template <int F, bool PROF, bool DBUF>
 __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
@@ -89,8 +89,11 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   constexpr int KS = F / 32;             // MFMA K-steps per dot product
   constexpr int IFR = TM_CHUNK / 16;     // item fragments per chunk
   extern __shared__ char lds_raw[];
+  // DBUF: two ys buffers, ONE barrier per chunk (compute buf[i&1] while
+  // draining the next chunk into buf[(i+1)&1])
   unsigned short* ys = reinterpret_cast<unsigned short*>(lds_raw);
-  float* topv = reinterpret_cast<float*>(lds_raw + TM_CHUNK * ROWB);
+  float* topv = reinterpret_cast<float*>(
+      lds_raw + (DBUF ? 2 : 1) * TM_CHUNK * ROWB);
   const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
   int* topi = reinterpret_cast<int*>(topv + TM_UPB * KP);
   float* th_lds = reinterpret_cast<float*>(topi + TM_UPB * KP);
@@ -154,46 +157,51 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   constexpr int NG = (TM_CHUNK * ROWB) / 16 / 256;  // granules per thread
   static_assert(NG >= 1, "chunk must cover one granule per thread");
   u32x4 stg[NG];
-  {
-    const long long base = it0;
+  auto load_stg = [&](long long cbase) {
 #pragma unroll
     for (int r = 0; r < NG; ++r) {
       const int lin = (tid + r * 256) * 16;
       const int row = lin / ROWB;
       const int col = lin % ROWB;
-      stg[r] = (base + row < it1)
-          ? *reinterpret_cast<const u32x4*>(&Y[(base + row) * F + col / 2])
+      stg[r] = (cbase + row < it1)
+          ? *reinterpret_cast<const u32x4*>(
+                &Y[(cbase + row) * F + col / 2])
           : u32x4{0u, 0u, 0u, 0u};
     }
-  }
-
-  for (long long base = it0; base < it1; base += TM_CHUNK) {
-    __syncthreads();  // all waves done reading ys from the previous chunk
-    // drain the in-flight registers into LDS (swizzled)
+  };
+  auto drain_to = [&](unsigned short* yb) {
 #pragma unroll
     for (int r = 0; r < NG; ++r) {
       const int lin = (tid + r * 256) * 16;
       const int row = lin / ROWB;
       const int col = lin % ROWB;
       const int dst = row * ROWB + (col ^ ((row & SWM) << 4));
-      *reinterpret_cast<u32x4*>(reinterpret_cast<char*>(ys) + dst) = stg[r];
+      *reinterpret_cast<u32x4*>(reinterpret_cast<char*>(yb) + dst) = stg[r];
     }
+  };
+  auto ybuf = [&](int i) -> unsigned short* {
+    return ys + (DBUF ? (size_t)(i & 1) * TM_CHUNK * F : 0);
+  };
+  const int n_chunks = (int)((it1 - it0 + TM_CHUNK - 1) / TM_CHUNK);
+  load_stg(it0);
+  if (DBUF) {
+    // prologue fill of buffer 0; chunk 1's loads fly under chunk 0
+    drain_to(ybuf(0));
     __syncthreads();
-    // issue the NEXT chunk's global loads here — after the barrier —
-    // so they fly under the MFMA phase and are only awaited by the
-    // next iteration's first barrier
-    const long long nbase = base + TM_CHUNK;
-    if (nbase < it1) {
-#pragma unroll
-      for (int r = 0; r < NG; ++r) {
-        const int lin = (tid + r * 256) * 16;
-        const int row = lin / ROWB;
-        const int col = lin % ROWB;
-        stg[r] = (nbase + row < it1)
-            ? *reinterpret_cast<const u32x4*>(
-                  &Y[(nbase + row) * F + col / 2])
-            : u32x4{0u, 0u, 0u, 0u};
-      }
+    if (n_chunks > 1) load_stg(it0 + TM_CHUNK);
+  }
+
+  for (int ci = 0; ci < n_chunks; ++ci) {
+    const long long base = it0 + (long long)ci * TM_CHUNK;
+    unsigned short* yb = ybuf(ci);
+    if (!DBUF) {
+      __syncthreads();  // all waves done reading the previous chunk
+      drain_to(yb);
+      __syncthreads();
+      // issue the NEXT chunk's global loads AFTER the barrier (a
+      // __syncthreads compiles to s_waitcnt vmcnt(0)) so they fly
+      // under the MFMA phase
+      if (base + TM_CHUNK < it1) load_stg(base + TM_CHUNK);
     }
     if (probe) {
       const unsigned long long now = wall_clock64();
@@ -212,7 +220,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
         const int row = i * 16 + lq;           // A operand row = item
         const int col = ks * 64 + lg * 16;     // byte offset of k-slice
         const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const char*>(ys) +
+            reinterpret_cast<const char*>(yb) +
             row * ROWB + (col ^ ((row & SWM) << 4)));
         acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, xf[ks], acc[i], 0, 0, 0);
@@ -273,6 +281,20 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
       acc_score += now - pt;
       pt = now;
     }
+    if (DBUF) {
+      // drain chunk ci+1 into the other buffer (its loads were issued
+      // one iteration ago), start chunk ci+2's loads, ONE barrier
+      if (ci + 1 < n_chunks) {
+        drain_to(ybuf(ci + 1));
+        if (ci + 2 < n_chunks) load_stg(base + 2 * TM_CHUNK);
+      }
+      __syncthreads();
+      if (probe) {
+        const unsigned long long now = wall_clock64();
+        acc_stage += now - pt;
+        pt = now;
+      }
+    }
   }
   __syncthreads();
   if (probe) pt = wall_clock64();
@@ -306,31 +328,48 @@ extern "C" void launch_topk_mfma(
 {
   dim3 grid((B + TM_UPB - 1) / TM_UPB, n_slices);
   dim3 block(256);
+  static const bool use_db = [] {
+    const char* e = getenv("PIO_TOPK_DB");
+    return e != nullptr && e[0] == '1';
+  }();
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
-    size_t lds_bytes = (size_t)TM_CHUNK * (FF * 2) +                         \
+    size_t lds_bytes = (size_t)(use_db ? 2 : 1) * TM_CHUNK * (FF * 2) +      \
                        (sizeof(float) + sizeof(int)) * TM_UPB * (K + 1) +    \
                        sizeof(float) * TM_UPB;                               \
     static bool attr_set_##FF = false;                                       \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                           \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, false>),       \
+          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, false, false>),       \
           hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, true>),        \
+          reinterpret_cast<const void*>(&topk_mfma_kernel<FF, true, false>),        \
           hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
       attr_set_##FF = true;                                                  \
     }                                                                        \
-    if (prof != nullptr)                                                     \
-      hipLaunchKernelGGL((topk_mfma_kernel<FF, true>), grid, block,          \
-                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
-                         ban_indices, out_val, out_idx, B, N, K, n_slices,   \
-                         item_base, prof);                                   \
-    else                                                                     \
-      hipLaunchKernelGGL((topk_mfma_kernel<FF, false>), grid, block,         \
-                         lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
-                         ban_indices, out_val, out_idx, B, N, K, n_slices,   \
-                         item_base, nullptr);                                \
+    if (prof != nullptr) {                                                   \
+      if (use_db)                                                            \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, true, true>), grid, block,  \
+                           lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
+                           ban_indices, out_val, out_idx, B, N, K, n_slices, \
+                           item_base, prof);                                 \
+      else                                                                   \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false>), grid, block, \
+                           lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
+                           ban_indices, out_val, out_idx, B, N, K, n_slices, \
+                           item_base, prof);                                 \
+    } else {                                                                 \
+      if (use_db)                                                            \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, false, true>), grid, block, \
+                           lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
+                           ban_indices, out_val, out_idx, B, N, K, n_slices, \
+                           item_base, nullptr);                              \
+      else                                                                   \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false>), grid,       \
+                           block, lds_bytes, stream, Xq, Y, item_mask,       \
+                           ban_indptr, ban_indices, out_val, out_idx, B, N,  \
+                           K, n_slices, item_base, nullptr);                 \
+    }                                                                        \
   } while (0)
   switch (f) {
     case 32: LAUNCH_M(32); break;

@@ -6,7 +6,8 @@ template's real train path — columnar bulk read -> factorize ID
 compaction -> device CSR -> fused ALS — and report events/s + s/iter.
 
   python scripts/train_from_store_bench.py --events 20000000   # CPU read
-  (on a GPU box: full train; on CPU: --read-only to skip the solve)
+  (on a GPU box: full train; on CPU: --read-only to skip the solve;
+   --tmpdir X --skip-seed reuses an already-seeded store)
 """
 import argparse
 import os
@@ -16,6 +17,38 @@ import time
 from datetime import datetime, timedelta, timezone
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+T0 = datetime(2020, 1, 1, tzinfo=timezone.utc)
+
+
+def seed(args, le, aid):
+    from predictionio_amd.data.events import DataMap, Event
+    import random
+    print(f"seeding {args.events:,} rate events "
+          f"({args.users:,} users x {args.items:,} items)...", flush=True)
+    ts = time.time()
+    rng = random.Random(7)
+    batch = 50_000
+    evs = [Event(event="rate", entity_type="user", entity_id="",
+                 target_entity_type="item", target_entity_id="",
+                 properties=DataMap({"rating": 3.0}), event_time=T0)
+           for _ in range(batch)]
+    done = 0
+    while done < args.events:
+        n = min(batch, args.events - done)
+        for j in range(n):
+            e = evs[j]
+            e.entity_id = f"u{rng.randrange(args.users)}"
+            e.target_entity_id = f"i{rng.randrange(args.items)}"
+            e.properties = DataMap({"rating": float(rng.randrange(1, 6))})
+            e.event_time = T0 + timedelta(seconds=done + j)
+            e.event_id = None
+        le.insert_batch(evs[:n], aid)
+        done += n
+        if done % 2_000_000 == 0:
+            print(f"  {done:,} ({done / (time.time() - ts):,.0f} ev/s)",
+                  flush=True)
+    print(f"seeded in {time.time() - ts:.1f}s", flush=True)
 
 
 def main():
@@ -28,9 +61,12 @@ def main():
     ap.add_argument("--read-only", action="store_true",
                     help="stop after the store->tensors ingest (no GPU)")
     ap.add_argument("--tmpdir", default=None)
+    ap.add_argument("--skip-seed", action="store_true",
+                    help="reuse an already-seeded store in --tmpdir")
     args = ap.parse_args()
 
     tmp = args.tmpdir or tempfile.mkdtemp(prefix="pio_store_bench")
+    print(f"store dir: {tmp}", flush=True)
     os.environ["PIO_STORAGE_SOURCES_BENCH_TYPE"] = "sqlite"
     os.environ["PIO_STORAGE_SOURCES_BENCH_PATH"] = os.path.join(
         tmp, "trainbench.sqlite")
@@ -38,44 +74,21 @@ def main():
         os.environ[f"PIO_STORAGE_REPOSITORIES_{repo}_SOURCE"] = "BENCH"
         os.environ[f"PIO_STORAGE_REPOSITORIES_{repo}_NAME"] = "bench"
     from predictionio_amd.data import storage
-    from predictionio_amd.data.events import DataMap, Event
     from predictionio_amd.data.storage.base import App
     storage.reset()
 
     apps = storage.get_meta_data_apps()
-    aid = apps.insert(App(0, "TrainBench", ""))
+    existing = apps.get_by_name("TrainBench")
+    if existing is not None:
+        aid = existing.id
+    else:
+        aid = apps.insert(App(0, "TrainBench", ""))
     le = storage.get_l_events()
     le.init(aid)
-
-    # ---- seed (setup, not the measured path — production ingest is the
-    # event server; ingest_bench.py measures that)
-    t0 = datetime(2020, 1, 1, tzinfo=timezone.utc)
-    print(f"seeding {args.events:,} rate events "
-          f"({args.users:,} users x {args.items:,} items)...", flush=True)
-    ts = time.time()
-    import random
-    rng = random.Random(7)
-    batch = 50_000
-    evs = [Event(event="rate", entity_type="user", entity_id="",
-                 target_entity_type="item", target_entity_id="",
-                 properties=DataMap({"rating": 3.0}), event_time=t0)
-           for _ in range(batch)]
-    done = 0
-    while done < args.events:
-        n = min(batch, args.events - done)
-        for j in range(n):
-            e = evs[j]
-            e.entity_id = f"u{rng.randrange(args.users)}"
-            e.target_entity_id = f"i{rng.randrange(args.items)}"
-            e.properties = DataMap({"rating": float(rng.randrange(1, 6))})
-            e.event_time = t0 + timedelta(seconds=done + j)
-            e.event_id = None
-        le.insert_batch(evs[:n], aid)
-        done += n
-        if done % 2_000_000 == 0:
-            print(f"  {done:,} ({done / (time.time() - ts):,.0f} ev/s)",
-                  flush=True)
-    print(f"seeded in {time.time() - ts:.1f}s", flush=True)
+    if not args.skip_seed:
+        # setup, not the measured path — production ingest is the event
+        # server; ingest_bench.py measures that
+        seed(args, le, aid)
 
     # ---- measured: the template's real train path from the store
     from predictionio_amd.controller import EngineParams, Params
@@ -116,7 +129,8 @@ def main():
         print(f"COMPACT+CSR: {n:,} events in {t4 - t3:.2f}s = "
               f"{n / (t4 - t3):,.0f} events/s (nnz={ix.numel():,}, "
               f"{len(uu):,} users x {len(ui):,} items)", flush=True)
-        print(f"TOTAL store->CSR: {n / (t4 - t1):,.0f} events/s", flush=True)
+        print(f"TOTAL store->CSR: {n / (t4 - t1):,.0f} events/s",
+              flush=True)
         return
     models = e.train(ep)
     t5 = time.time()
