@@ -243,25 +243,51 @@ class CooccurrenceAlgorithm(Algorithm):
     (CooccurrenceAlgorithm.scala:49-108)."""
 
     def train(self, pd: PreparedData) -> CooccurrenceModel:
+        """Pair counting as ONE sparse Gramian: C = A^T A with A the
+        user×item view-incidence matrix (user-distinct), then a
+        vectorized per-item top-N — replaces the reference's self-join
+        + groupBy (CooccurrenceAlgorithm.scala:55-108) and round 1's
+        Python pair loops (quadratic in per-user views). Runs on the
+        GPU when available (torch sparse mm)."""
+        import numpy as np
         item_map = BiMap.string_int(
             [v.item for v in pd.view_events] + list(pd.items))
+        user_map = BiMap.string_int(v.user for v in pd.view_events)
         n_keep = int(self.params.get("n", 10))
-        # user-distinct item sets → pair counts
-        by_user: Dict[str, Set[int]] = defaultdict(set)
-        for v in pd.view_events:
-            by_user[v.user].add(item_map[v.item])
-        pair_counts: Dict[Tuple[int, int], int] = defaultdict(int)
-        for items in by_user.values():
-            s = sorted(items)
-            for a_i, a in enumerate(s):
-                for b in s[a_i + 1:]:
-                    pair_counts[(a, b)] += 1
-        per_item: Dict[int, List[Tuple[int, int]]] = defaultdict(list)
-        for (a, b), c in pair_counts.items():
-            per_item[a].append((b, c))
-            per_item[b].append((a, c))
-        top_n = {i: sorted(l, key=lambda t: -t[1])[:n_keep]
-                 for i, l in per_item.items()}
+        n_items = len(item_map)
+        n_users = len(user_map)
+        if not pd.view_events:
+            return CooccurrenceModel({}, item_map, pd.items)
+        u = torch.tensor([user_map[v.user] for v in pd.view_events],
+                         dtype=torch.int64)
+        i = torch.tensor([item_map[v.item] for v in pd.view_events],
+                         dtype=torch.int64)
+        # user-distinct (view twice ≠ two pairs)
+        key = u * n_items + i
+        key = torch.unique(key)
+        u, i = key // n_items, key % n_items
+        device = (torch.device("cuda") if torch.cuda.is_available()
+                  else torch.device("cpu"))
+        A = torch.sparse_coo_tensor(
+            torch.stack([u, i]).to(device),
+            torch.ones(u.numel(), device=device),
+            (n_users, n_items)).coalesce()
+        C = torch.sparse.mm(A.t(), A).coalesce()
+        ii = C.indices()[0].cpu().numpy()
+        jj = C.indices()[1].cpu().numpy()
+        vv = C.values().cpu().numpy()
+        off = ii != jj  # drop the diagonal (item with itself)
+        ii, jj, vv = ii[off], jj[off], vv[off].astype(np.int64)
+        # per-item top-N: lexsort by (item, -count), take first n_keep
+        order = np.lexsort((-vv, ii))
+        ii, jj, vv = ii[order], jj[order], vv[order]
+        starts = np.flatnonzero(np.r_[True, ii[1:] != ii[:-1]])
+        top_n: Dict[int, List[Tuple[int, int]]] = {}
+        bounds = np.r_[starts, len(ii)]
+        for s, e in zip(bounds[:-1], bounds[1:]):
+            e2 = min(e, s + n_keep)
+            top_n[int(ii[s])] = list(zip(jj[s:e2].tolist(),
+                                         vv[s:e2].tolist()))
         return CooccurrenceModel(top_n, item_map, pd.items)
 
     def predict(self, model: CooccurrenceModel, query) -> PredictedResult:
