@@ -157,16 +157,30 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   constexpr int NG = (TM_CHUNK * ROWB) / 16 / 256;  // granules per thread
   static_assert(NG >= 1, "chunk must cover one granule per thread");
   u32x4 stg[NG];
-  auto load_stg = [&](long long cbase) {
+  // per-thread element offsets are chunk-invariant: loads use ONE
+  // running base pointer advanced by a constant per chunk (the naive
+  // form re-did ~20 64-bit address ops per chunk in the hot loop)
+  int goff[NG];
+  int grow[NG];
 #pragma unroll
-    for (int r = 0; r < NG; ++r) {
-      const int lin = (tid + r * 256) * 16;
-      const int row = lin / ROWB;
-      const int col = lin % ROWB;
-      stg[r] = (cbase + row < it1)
-          ? *reinterpret_cast<const u32x4*>(
-                &Y[(cbase + row) * F + col / 2])
-          : u32x4{0u, 0u, 0u, 0u};
+  for (int r = 0; r < NG; ++r) {
+    const int lin = (tid + r * 256) * 16;
+    grow[r] = lin / ROWB;
+    goff[r] = grow[r] * F + (lin % ROWB) / 2;
+  }
+  auto load_stg = [&](long long cbase) {
+    const unsigned short* yb_g = Y + cbase * F;
+    const bool tail = cbase + TM_CHUNK > it1;
+    if (!tail) {
+#pragma unroll
+      for (int r = 0; r < NG; ++r)
+        stg[r] = *reinterpret_cast<const u32x4*>(&yb_g[goff[r]]);
+    } else {
+#pragma unroll
+      for (int r = 0; r < NG; ++r)
+        stg[r] = (cbase + grow[r] < it1)
+            ? *reinterpret_cast<const u32x4*>(&yb_g[goff[r]])
+            : u32x4{0u, 0u, 0u, 0u};
     }
   };
   auto drain_to = [&](unsigned short* yb) {
@@ -249,13 +263,18 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
         asm volatile("" ::: "memory");  // keep iterations distinct
         if (lg == g) {
           float th = th_lds[mylist];
+          // 32-bit tail guard: li < lim replaces the former 64-bit
+          // `item < it1` (16 hoisted v_cmp_gt_i64 in the fast loop)
+          const int lim = (int)(it1 - base < TM_CHUNK ? it1 - base
+                                                      : TM_CHUNK);
 #pragma unroll
           for (int i = 0; i < IFR; ++i) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-              const long long item = base + i * 16 + lg * 4 + r;  // D row
+              const int li = i * 16 + lg * 4 + r;      // D row
+              const long long item = base + li;
               const float s = acc[i][r];
-              if (s > th && item < it1) {
+              if (s > th && li < lim) {
                 if ((item_mask == nullptr || !item_mask[item]) &&
                     (ban == nullptr ||
                      !tm_in_sorted(ban, bn, (int)(item + item_base)))) {
