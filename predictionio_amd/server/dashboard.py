@@ -33,11 +33,47 @@ def create_app() -> FastAPI:
             f"<td><a href='/engine_instances/{i.id}/evaluator_results.json'>"
             "JSON</a></td></tr>"
             for i in insts)
+        # training runs (beyond the reference's eval-only dashboard:
+        # the EngineInstance records CoreWorkflow writes)
+        tr = sorted(storage.get_meta_data_engine_instances().get_all(),
+                    key=lambda i: i.start_time, reverse=True)[:50]
+        trows = "".join(
+            f"<tr><td><a href='/training/{i.id}'>{i.id}</a></td>"
+            f"<td>{escape(i.status)}</td>"
+            f"<td>{escape(i.engine_factory)}</td>"
+            f"<td>{escape(i.engine_variant)}</td>"
+            f"<td>{i.start_time}</td><td>{i.end_time}</td></tr>"
+            for i in tr)
         return f"""<html><head><title>PredictionIO-AMD Dashboard</title>
 </head><body><h1>Completed Evaluations</h1>
 <table border=1 cellpadding=4>
 <tr><th>ID</th><th>Evaluation</th><th>Generator</th><th>Start</th>
-<th>End</th><th>Results</th></tr>{rows}</table></body></html>"""
+<th>End</th><th>Results</th></tr>{rows}</table>
+<h1>Training Runs</h1>
+<table border=1 cellpadding=4>
+<tr><th>ID</th><th>Status</th><th>Engine factory</th><th>Variant</th>
+<th>Start</th><th>End</th></tr>{trows}</table></body></html>"""
+
+    @app.get("/training/{iid}", response_class=HTMLResponse)
+    def training_detail(iid: str):
+        i = storage.get_meta_data_engine_instances().get(iid)
+        if i is None:
+            return HTMLResponse("<h1>Not Found</h1>", status_code=404)
+        fields = {
+            "status": i.status, "engineFactory": i.engine_factory,
+            "engineVariant": i.engine_variant, "batch": i.batch,
+            "startTime": str(i.start_time), "endTime": str(i.end_time),
+            "dataSourceParams": i.data_source_params,
+            "preparatorParams": i.preparator_params,
+            "algorithmsParams": i.algorithms_params,
+            "servingParams": i.serving_params,
+        }
+        rows = "".join(f"<tr><td>{escape(k)}</td>"
+                       f"<td><pre>{escape(str(v))}</pre></td></tr>"
+                       for k, v in fields.items())
+        return (f"<html><body><h1>Training {i.id}</h1>"
+                f"<table border=1 cellpadding=4>{rows}</table>"
+                "</body></html>")
 
     @app.get("/engine_instances/{iid}", response_class=HTMLResponse)
     def detail(iid: str):

@@ -214,3 +214,55 @@ class TestDashboardAdmin:
         assert c.delete("/cmd/app/adminapp/data").status_code == 200
         assert c.delete("/cmd/app/adminapp").status_code == 200
         assert c.delete("/cmd/app/adminapp").status_code == 404
+
+
+class TestAdminExtended:
+    def test_keys_channels_status(self, mem_storage):
+        from fastapi.testclient import TestClient
+
+        from predictionio_amd.server.admin import create_app
+        c = TestClient(create_app())
+        assert c.get("/status").json()["status"] == "ok"
+        r = c.post("/cmd/app", json={"name": "extapp"})
+        assert r.status_code == 200
+        # access keys
+        k = c.post("/cmd/app/extapp/accesskey",
+                   json={"events": ["view"]}).json()["accessKey"]
+        detail = c.get("/cmd/app/extapp").json()
+        assert any(e["key"] == k and e["events"] == ["view"]
+                   for e in detail["accessKeys"])
+        assert c.delete(f"/cmd/accesskey/{k}").status_code == 200
+        assert c.delete(f"/cmd/accesskey/{k}").status_code == 404
+        # channels
+        r = c.post("/cmd/app/extapp/channel", json={"name": "live"})
+        assert r.status_code == 200
+        assert c.post("/cmd/app/extapp/channel",
+                      json={"name": "bad name!"}).status_code == 400
+        assert any(ch["name"] == "live"
+                   for ch in c.get("/cmd/app/extapp").json()["channels"])
+        assert c.delete("/cmd/app/extapp/channel/live").status_code == 200
+        assert c.delete("/cmd/app/extapp/channel/live").status_code == 404
+        # engine instances listing
+        assert c.get("/cmd/engineinstances").json()["engineInstances"] == []
+
+
+class TestDashboardTraining:
+    def test_training_runs_listed(self, mem_storage):
+        from fastapi.testclient import TestClient
+
+        from predictionio_amd.data.storage.base import (
+            EngineInstance, utcnow,
+        )
+        from predictionio_amd.server.dashboard import create_app
+        ei = mem_storage.get_meta_data_engine_instances()
+        iid = ei.insert(EngineInstance(
+            id="", status="COMPLETED", start_time=utcnow(),
+            end_time=utcnow(), engine_id="e", engine_version="1",
+            engine_variant="engine.json", engine_factory="MyFactory",
+            algorithms_params="[('als', {...})]"))
+        c = TestClient(create_app())
+        html = c.get("/").text
+        assert "MyFactory" in html and iid in html
+        detail = c.get(f"/training/{iid}").text
+        assert "MyFactory" in detail and "algorithmsParams" in detail
+        assert c.get("/training/nope").status_code == 404
