@@ -42,6 +42,9 @@ def woodbury_lv(Y: torch.Tensor, YtY: torch.Tensor,
     # while hipBLASLt GEMMs of that shape are proven.
     Linv = torch.linalg.solve_triangular(L, eye, upper=False)
     V = (Y @ Linv.mT).contiguous()  # row v_i = L^-1 y_i
+    import os as _os
+    if (_os.environ.get("PIO_ALS_STAGE_BF16") == "1" and Y.is_cuda):
+        V = V.to(torch.bfloat16).contiguous()
     return Linv, V
 
 
@@ -108,6 +111,13 @@ def als_solve(indptr: torch.Tensor, indices: torch.Tensor,
                 Linv, V = lv
             else:
                 Linv, V = woodbury_lv(Yp, YtYp, lam)
+            # opt-in bf16 staging of the whitened factors (see the
+            # BF16S kernel variant; numerics bounded by
+            # profiles/bf16_numerics_study.txt)
+            import os as _os
+            if _os.environ.get("PIO_ALS_STAGE_BF16") == "1" \
+                    and V.dtype == torch.float32:
+                V = V.to(torch.bfloat16).contiguous()
             Z = ext.als_solve(ip, ix, vv, Yp, YtYp, V, float(lam),
                               float(alpha), True, False, 1, None)
             # X = Z L^-1 for all rows (Woodbury rows hold z; big rows get

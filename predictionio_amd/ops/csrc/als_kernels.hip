@@ -551,7 +551,14 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
 // smaller LDS footprint (9.7 vs 13.3 KB/wave) fits 16 waves/CU instead of
 // 12 and whose unrolled M-solve is 25% shorter — the phase probe showed
 // the kernel issue-bound at its occupancy cap with the M-solve at 66%.
-template <int F, int NW, int NLO>
+// BF16S: stage the factor rows (V in implicit mode) as bf16 — halves the
+// occupancy-capping Yl LDS buffer AND the stage HBM bytes. The round-2
+// TCC probe showed this kernel latency-bound (SQ_WAIT:BUSY 19:1) at only
+// ~310 GB/s effective, so resident waves are the lever; the bf16
+// numerics study (profiles/bf16_numerics_study.txt) bounds the staged-
+// factor rounding at ~1e-5 relative ALS objective. Opt-in via
+// PIO_ALS_STAGE_BF16=1 (the V pointer then carries bf16 data).
+template <int F, int NW, int NLO, bool BF16S = false>
 __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
@@ -570,18 +577,21 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
                                      // so the kernel self-times sampled
                                      // rows with wall_clock64()
 {
-  constexpr int FP = F + 4;  // bank-group padding
+  constexpr int FP = F + 4;    // fp32 row stride (floats)
+  constexpr int FPH = F + 8;   // bf16 row stride (ushorts; 16B multiple)
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  __shared__ float Yl[2][NW][FP];      // staged factor rows (Y or V)
+  __shared__ float Yl[2][NW][BF16S ? (FPH / 2) : FP];  // staged rows
   constexpr int MP = NW + 4;  // M row stride: 16B-aligned for b128 reads
   __shared__ float M[2][NW][MP];       // I+DGD (implicit) / G+regI (explicit)
   __shared__ float tv[2][NW];          // col ids, then rhs, then solution
   __shared__ float dv[2][NW];          // D diagonal (implicit)
 
   float* yl = &Yl[wave][0][0];
+  unsigned short* ylh = reinterpret_cast<unsigned short*>(yl);
   const float* src = implicit_mode ? V : Y;
+  const unsigned short* srch = reinterpret_cast<const unsigned short*>(V);
 
   for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
        row += (long long)gridDim.x * 2) {
@@ -609,8 +619,16 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     wave_sync();
     for (int c = 0; c < n; ++c) {
       const long long col = __float_as_int(tv[wave][c]);
-      for (int e = lane; e < F; e += 64)
-        yl[c * FP + e] = src[col * F + e];
+      if constexpr (BF16S) {
+        const unsigned int* srow = reinterpret_cast<const unsigned int*>(
+            srch + col * F);
+        unsigned int* drow =
+            reinterpret_cast<unsigned int*>(ylh + c * FPH);
+        for (int e = lane; e < F / 2; e += 64) drow[e] = srow[e];
+      } else {
+        for (int e = lane; e < F; e += 64)
+          yl[c * FP + e] = src[col * F + e];
+      }
     }
     if (lane < n) {
       const float r = values[start + lane];
@@ -640,12 +658,36 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
       while (p >= n - i && i < n) { p -= n - i; ++i; }
       int j = i + p;
       for (int pp = lane; pp < npairs; pp += 64) {
-        const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
-        const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
-        f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};  // 2x v_pk_fma_f32 per q
+        float dot;
+        if constexpr (BF16S) {
+          // b128 reads of 8 bf16; bf16->f32 is a 16-bit shift/mask on
+          // the packed u32 (the bf16 bit pattern IS the f32 high half)
+          const uint4* yi8 = reinterpret_cast<const uint4*>(&ylh[i * FPH]);
+          const uint4* yj8 = reinterpret_cast<const uint4*>(&ylh[j * FPH]);
+          f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
-        float dot = acc4.x + acc4.y + acc4.z + acc4.w;
+          for (int q = 0; q < F / 8; ++q) {
+            const uint4 a = yi8[q];
+            const uint4 b = yj8[q];
+            const unsigned int av[4] = {a.x, a.y, a.z, a.w};
+            const unsigned int bv[4] = {b.x, b.y, b.z, b.w};
+#pragma unroll
+            for (int t = 0; t < 4; ++t) {
+              acc4.x = fmaf(__uint_as_float(av[t] << 16),
+                            __uint_as_float(bv[t] << 16), acc4.x);
+              acc4.y = fmaf(__uint_as_float(av[t] & 0xffff0000u),
+                            __uint_as_float(bv[t] & 0xffff0000u), acc4.y);
+            }
+          }
+          dot = acc4.x + acc4.y;
+        } else {
+          const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
+          const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
+          f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};  // 2x v_pk_fma_f32 per q
+#pragma unroll
+          for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
+          dot = acc4.x + acc4.y + acc4.z + acc4.w;
+        }
         if (implicit_mode) {
           dot *= dv[wave][i] * dv[wave][j];
           if (i == j) dot += 1.f;
@@ -737,8 +779,14 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
     //      or   sum_i s_i y_i (explicit: x directly) ----
     for (int e = lane; e < F; e += 64) {
       float x = 0.f;
-      for (int c = 0; c < n; ++c)
-        x = fmaf(tv[wave][c], yl[c * FP + e], x);
+      for (int c = 0; c < n; ++c) {
+        float ye;
+        if constexpr (BF16S)
+          ye = __uint_as_float((unsigned int)ylh[c * FPH + e] << 16);
+        else
+          ye = yl[c * FP + e];
+        x = fmaf(tv[wave][c], ye, x);
+      }
       X[row * (long long)F + e] = x;
     }
     if (probe) {
@@ -1017,6 +1065,13 @@ extern "C" void launch_als_solve(
     const char* e = getenv("PIO_ALS_DUAL");
     return e != nullptr && e[0] == '1';
   }();
+  // PIO_ALS_STAGE_BF16=1: V arrives as bf16 (the Python side casts) and
+  // the Woodbury kernels stage it as bf16 — implicit mode only
+  static const bool stage_bf16 = [] {
+    const char* e = getenv("PIO_ALS_STAGE_BF16");
+    return e != nullptr && e[0] == '1';
+  }();
+  const bool use_bf16s = stage_bf16 && implicit_mode && V != nullptr;
   long long wg4 = ((long long)n_rows + 3) / 4;
   int grid_w2 = (int)(wg4 < (1 << 20) ? wg4 : (1 << 20));
 #define LAUNCH_WOODBURY(FF)                                                  \
@@ -1036,6 +1091,19 @@ extern "C" void launch_als_solve(
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \
                        X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
                        prof);                                                \
+  } else if (woodbury && use_bf16s) {                                        \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 20, -1, true>),              \
+                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
+                       values, Y, V, X, n_rows, lambda, alpha,               \
+                       implicit_mode, wr_scale, prof);                       \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 24, 20, true>),              \
+                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
+                       values, Y, V, X, n_rows, lambda, alpha,               \
+                       implicit_mode, wr_scale, prof);                       \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24, true>),              \
+                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
+                       values, Y, V, X, n_rows, lambda, alpha,               \
+                       implicit_mode, wr_scale, prof);                       \
   } else if (woodbury) {                                                     \
     hipLaunchKernelGGL((als_woodbury_kernel<FF, 20, -1>), dim3(grid_w),      \
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \

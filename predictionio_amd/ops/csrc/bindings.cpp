@@ -77,9 +77,20 @@ torch::Tensor als_solve(torch::Tensor indptr, torch::Tensor indices,
   }
   const float* v_ptr = nullptr;
   if (V.has_value()) {
-    check_cuda_f32(*V, "V");
+    // V may be bf16 when PIO_ALS_STAGE_BF16 is set (the kernel stages
+    // it as bf16; the pointer is reinterpreted device-side)
+    TORCH_CHECK(V->is_cuda() && V->is_contiguous() &&
+                    (V->scalar_type() == torch::kFloat32 ||
+                     V->scalar_type() == torch::kBFloat16),
+                "V must be contiguous fp32/bf16 GPU");
     TORCH_CHECK(V->sizes() == Y.sizes(), "V must match Y shape");
-    v_ptr = V->data_ptr<float>();
+    const bool env_bf16 = [] {
+      const char* e = getenv("PIO_ALS_STAGE_BF16");
+      return e != nullptr && e[0] == '1';
+    }();
+    TORCH_CHECK((V->scalar_type() == torch::kBFloat16) == env_bf16,
+                "V dtype must match PIO_ALS_STAGE_BF16");
+    v_ptr = reinterpret_cast<const float*>(V->data_ptr());
   }
   unsigned long long* prof_ptr = nullptr;
   if (prof.has_value()) {
