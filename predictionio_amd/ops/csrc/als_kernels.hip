@@ -812,7 +812,7 @@ __global__ __launch_bounds__(128, 6) void als_woodbury_kernel(
 // row, so their per-row cost is unchanged.
 // ---------------------------------------------------------------------------
 
-template <int F, int NW, int NLO>
+template <int F, int NW, int NLO, bool BF16S = false>
 __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
@@ -828,18 +828,21 @@ __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
     unsigned long long* prof)
 {
   constexpr int FP = F + 4;
+  constexpr int FPH = F + 8;  // bf16 row stride (ushorts)
   constexpr int MP = NW + 4;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int h = lane >> 5;    // wave half = which of the two rows
   const int r = lane & 31;    // sub-lane within the half
 
-  __shared__ float Yl[2][2][NW][FP];   // [wave][half]
+  __shared__ float Yl[2][2][NW][BF16S ? (FPH / 2) : FP];  // [wave][half]
+  constexpr int YROW = BF16S ? (FPH / 2) : FP;  // floats per staged row
   __shared__ float M[2][2][NW][MP];
   __shared__ float tv[2][2][NW];
   __shared__ float dv[2][2][NW];
 
   const float* src = implicit_mode ? V : Y;
+  const unsigned short* srch = reinterpret_cast<const unsigned short*>(V);
 
   for (long long base = ((long long)blockIdx.x * 2 + wave) * 2;
        base < n_rows; base += (long long)gridDim.x * 4) {
@@ -878,13 +881,22 @@ __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
       const int n = nh[hh];
       const long long start = starth[hh];
       float* yl = &Yl[wave][hh][0][0];
+      unsigned short* ylh = reinterpret_cast<unsigned short*>(yl);
       if (lane < n)
         tv[wave][hh][lane] = __int_as_float(indices[start + lane]);
       wave_sync();
       for (int c = 0; c < n; ++c) {
         const long long col = __float_as_int(tv[wave][hh][c]);
-        for (int e = lane; e < F; e += 64)
-          yl[c * FP + e] = src[col * F + e];
+        if constexpr (BF16S) {
+          const unsigned int* srow = reinterpret_cast<const unsigned int*>(
+              srch + col * F);
+          unsigned int* drow =
+              reinterpret_cast<unsigned int*>(ylh + c * FPH);
+          for (int e = lane; e < F / 2; e += 64) drow[e] = srow[e];
+        } else {
+          for (int e = lane; e < F; e += 64)
+            yl[c * FP + e] = src[col * F + e];
+        }
       }
       if (lane < n) {
         const float v = values[start + lane];
@@ -905,18 +917,42 @@ __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
     const int n = nh[h];           // this half's row size (per-lane)
     {
       const float* yl = &Yl[wave][h][0][0];
+      const unsigned short* ylh =
+          reinterpret_cast<const unsigned short*>(yl);
       const float reg = wr_scale ? lambda * (float)n : lambda;
       const int npairs = act[h] ? n * (n + 1) / 2 : 0;
       int p = r, i = 0;
       while (p >= n - i && i < n) { p -= n - i; ++i; }
       int j = i + p;
       for (int pp = r; pp < npairs; pp += 32) {
-        const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
-        const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
-        f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};
+        float dot;
+        if constexpr (BF16S) {
+          const uint4* yi8 = reinterpret_cast<const uint4*>(&ylh[i * FPH]);
+          const uint4* yj8 = reinterpret_cast<const uint4*>(&ylh[j * FPH]);
+          f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
-        float dot = acc4.x + acc4.y + acc4.z + acc4.w;
+          for (int q = 0; q < F / 8; ++q) {
+            const uint4 a = yi8[q];
+            const uint4 b = yj8[q];
+            const unsigned int av[4] = {a.x, a.y, a.z, a.w};
+            const unsigned int bv[4] = {b.x, b.y, b.z, b.w};
+#pragma unroll
+            for (int t = 0; t < 4; ++t) {
+              acc4.x = fmaf(__uint_as_float(av[t] << 16),
+                            __uint_as_float(bv[t] << 16), acc4.x);
+              acc4.y = fmaf(__uint_as_float(av[t] & 0xffff0000u),
+                            __uint_as_float(bv[t] & 0xffff0000u), acc4.y);
+            }
+          }
+          dot = acc4.x + acc4.y;
+        } else {
+          const f32x4_t* yi = reinterpret_cast<const f32x4_t*>(&yl[i * FP]);
+          const f32x4_t* yj = reinterpret_cast<const f32x4_t*>(&yl[j * FP]);
+          f32x4_t acc4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+          for (int q = 0; q < F / 4; ++q) acc4 += yi[q] * yj[q];
+          dot = acc4.x + acc4.y + acc4.z + acc4.w;
+        }
         if (implicit_mode) {
           dot *= dv[wave][h][i] * dv[wave][h][j];
           if (i == j) dot += 1.f;
@@ -1000,12 +1036,21 @@ __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
     for (int hh = 0; hh < 2; ++hh) {
       if (!act[hh]) continue;
       const float* yl = &Yl[wave][hh][0][0];
+      const unsigned short* ylh =
+          reinterpret_cast<const unsigned short*>(yl);
       const int nn = nh[hh];
       const long long row = base + hh;
       for (int e = lane; e < F; e += 64) {
         float x = 0.f;
-        for (int c = 0; c < nn; ++c)
-          x = fmaf(tv[wave][hh][c], yl[c * FP + e], x);
+        for (int c = 0; c < nn; ++c) {
+          float ye;
+          if constexpr (BF16S)
+            ye = __uint_as_float(
+                (unsigned int)ylh[c * FPH + e] << 16);
+          else
+            ye = yl[c * FP + e];
+          x = fmaf(tv[wave][hh][c], ye, x);
+        }
         X[row * (long long)F + e] = x;
       }
     }
@@ -1075,7 +1120,20 @@ extern "C" void launch_als_solve(
   long long wg4 = ((long long)n_rows + 3) / 4;
   int grid_w2 = (int)(wg4 < (1 << 20) ? wg4 : (1 << 20));
 #define LAUNCH_WOODBURY(FF)                                                  \
-  if (woodbury && use_dual) {                                                \
+  if (woodbury && use_dual && use_bf16s) {                                   \
+    hipLaunchKernelGGL((als_woodbury2_kernel<FF, 20, -1, true>),             \
+                       dim3(grid_w2), dim3(128), 0, stream, indptr,          \
+                       indices, values, Y, V, X, n_rows, lambda, alpha,      \
+                       implicit_mode, wr_scale, prof);                       \
+    hipLaunchKernelGGL((als_woodbury2_kernel<FF, 24, 20, true>),             \
+                       dim3(grid_w2), dim3(128), 0, stream, indptr,          \
+                       indices, values, Y, V, X, n_rows, lambda, alpha,      \
+                       implicit_mode, wr_scale, prof);                       \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24, true>),              \
+                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
+                       values, Y, V, X, n_rows, lambda, alpha,               \
+                       implicit_mode, wr_scale, prof);                       \
+  } else if (woodbury && use_dual) {                                         \
     /* dual for nnz<=24; the NW=32 dual instantiation spills (mr[32] x   */  \
     /* dual-row bookkeeping exceeds the register budget) so rows 25-32   */  \
     /* keep the single-row kernel                                        */  \
