@@ -1106,16 +1106,12 @@ extern "C" void launch_als_solve(
                      lambda, alpha, implicit_mode, wr_scale, skip)
   // PIO_ALS_DUAL=1 routes the Woodbury rows through the dual-row kernel
   // (two rows per wave) — A/B lever for the round-2 M-solve rework
-  static const bool use_dual = [] {
-    const char* e = getenv("PIO_ALS_DUAL");
-    return e != nullptr && e[0] == '1';
-  }();
+  const char* e_dual = getenv("PIO_ALS_DUAL");
+  const bool use_dual = e_dual != nullptr && e_dual[0] == '1';
   // PIO_ALS_STAGE_BF16=1: V arrives as bf16 (the Python side casts) and
   // the Woodbury kernels stage it as bf16 — implicit mode only
-  static const bool stage_bf16 = [] {
-    const char* e = getenv("PIO_ALS_STAGE_BF16");
-    return e != nullptr && e[0] == '1';
-  }();
+  const char* e_bf16 = getenv("PIO_ALS_STAGE_BF16");
+  const bool stage_bf16 = e_bf16 != nullptr && e_bf16[0] == '1';
   const bool use_bf16s = stage_bf16 && implicit_mode && V != nullptr;
   long long wg4 = ((long long)n_rows + 3) / 4;
   int grid_w2 = (int)(wg4 < (1 << 20) ? wg4 : (1 << 20));

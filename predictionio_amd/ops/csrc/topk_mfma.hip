@@ -347,10 +347,8 @@ extern "C" void launch_topk_mfma(
 {
   dim3 grid((B + TM_UPB - 1) / TM_UPB, n_slices);
   dim3 block(256);
-  static const bool use_db = [] {
-    const char* e = getenv("PIO_TOPK_DB");
-    return e != nullptr && e[0] == '1';
-  }();
+  const char* e_db = getenv("PIO_TOPK_DB");
+  const bool use_db = e_db != nullptr && e_db[0] == '1';
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
     size_t lds_bytes = (size_t)(use_db ? 2 : 1) * TM_CHUNK * (FF * 2) +      \
