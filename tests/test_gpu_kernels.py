@@ -4,6 +4,7 @@ Run on an MI355X via gpurun: python -m pytest tests -m gpu -x -q
 """
 
 import math
+import os
 
 import pytest
 import torch
@@ -382,3 +383,39 @@ class TestEvalOnGPU:
         res = MetricEvaluator(PrecisionAtK(k=5)).evaluate_base(
             e, e.batch_eval([ep]))
         assert 0.0 <= res.best_score <= 1.0
+
+
+@requires_gpu
+class TestWoodburyOptIns:
+    def test_dual_bf16_combo_numerics(self):
+        """The opt-in PIO_ALS_DUAL + PIO_ALS_STAGE_BF16 path must stay
+        within the bf16-staging numerics bound (study: ~2e-3 relative).
+        Runs in a subprocess so the env flags apply from the first
+        launch."""
+        import subprocess
+        import sys
+        code = """
+import torch, math
+from predictionio_amd.ops import als as als_ops
+g = torch.Generator().manual_seed(3)
+n_rows, n_cols, f = 3000, 1500, 64
+nnz = n_rows * 15
+indptr = torch.arange(0, nnz + 1, 15, dtype=torch.int64)[: n_rows + 1]
+ix = torch.randint(0, n_cols, (nnz,), generator=g, dtype=torch.int32)
+vv = torch.ones(nnz)
+Y = (torch.randn((n_cols, f), generator=g) / math.sqrt(f)).float()
+YtY = als_ops.gramian(Y)
+ref = als_ops.als_solve_ref(indptr, ix, vv, Y, YtY, lam=0.01,
+                            alpha=40.0, implicit=True)
+X = als_ops.als_solve(indptr.cuda(), ix.cuda(), vv.cuda(), Y.cuda(),
+                      YtY.cuda(), lam=0.01, alpha=40.0,
+                      implicit=True).cpu()
+rel = ((X - ref).norm() / ref.norm()).item()
+assert rel < 5e-3, rel
+print("REL", rel)
+"""
+        env = dict(os.environ, PIO_ALS_DUAL="1", PIO_ALS_STAGE_BF16="1")
+        r = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stdout + r.stderr
+        assert "REL" in r.stdout
