@@ -353,7 +353,8 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
     float alpha,
     int implicit_mode,
     int wr_scale,
-    int skip_below)
+    int skip_below,
+    int skip_above)   // rows with nnz > this go to the workgroup kernel
 {
   static_assert(F <= 64, "wave kernel supports rank <= 64");
   const int lane = threadIdx.x & 63;
@@ -368,7 +369,8 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
        row += (long long)gridDim.x * 2) {
     const long long start = indptr[row];
     const int nnz = (int)(indptr[row + 1] - start);
-    if (nnz <= skip_below) continue;  // Woodbury kernel owns these rows
+    if (nnz <= skip_below || nnz > skip_above)
+      continue;  // Woodbury / workgroup kernels own these rows
 
     float acc[F];
 #pragma unroll
@@ -1094,11 +1096,23 @@ extern "C" void launch_als_solve(
       (implicit_mode ? V != nullptr : true);
   const int skip = (woodbury || which == 2) ? WOODBURY_MAX_NNZ : -1;
   const bool dense = which != 1;
+  // PIO_ALS_DENSE_SPLIT=N routes rows with nnz > N to the
+  // workgroup-per-row kernel (256 threads/row — 4x the per-row
+  // parallelism of the wave kernel for very dense rows, e.g. the item
+  // side at 8 GPUs where nnz ~ 200). 0/unset = wave kernel takes all.
+  const char* e_split = getenv("PIO_ALS_DENSE_SPLIT");
+  const int dense_split = e_split ? atoi(e_split) : 0;
+  const int wave_hi = dense_split > 0 ? dense_split : 0x7fffffff;
 #define LAUNCH_WAVE(FF)                                                      \
-  if (dense)                                                                 \
+  if (dense) {                                                               \
   hipLaunchKernelGGL((als_solve_wave_kernel<FF>), dim3(grid_w), dim3(128),   \
                      0, stream, indptr, indices, values, Y, YtY, X, n_rows,  \
-                     lambda, alpha, implicit_mode, wr_scale, skip)
+                     lambda, alpha, implicit_mode, wr_scale, skip, wave_hi); \
+  if (dense_split > 0)                                                       \
+    hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,   \
+                       stream, indptr, indices, values, Y, YtY, X, n_rows,   \
+                       lambda, alpha, implicit_mode, wr_scale, dense_split); \
+  }
 #define LAUNCH_BLOCK(FF)                                                     \
   if (dense)                                                                 \
   hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,     \
