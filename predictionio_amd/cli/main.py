@@ -254,6 +254,16 @@ def adminserver(ip, port):
 
 
 @cli.command()
+def unregister():
+    """Unregister an engine (Console.scala `unregister` — the verb is
+    parsed by the reference but engine registration was removed after
+    manifests were dropped; kept for CLI-surface parity)."""
+    click.echo("[INFO] Engine registration is no longer used; nothing "
+               "to unregister (parity with the reference's vestigial "
+               "verb).")
+
+
+@cli.command()
 @click.option("--ip", default="0.0.0.0")
 @click.option("--port", default=7072)
 def storageserver(ip, port):
