@@ -72,7 +72,7 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
-template <int F, bool PROF, bool DBUF>
+template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK>
 __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
@@ -87,13 +87,13 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   constexpr int ROWB = F * 2;            // bytes per staged Y row
   constexpr int SWM = (F >= 64) ? 7 : 3; // XOR-swizzle row mask
   constexpr int KS = F / 32;             // MFMA K-steps per dot product
-  constexpr int IFR = TM_CHUNK / 16;     // item fragments per chunk
+  constexpr int IFR = CH / 16;           // item fragments per chunk
   extern __shared__ char lds_raw[];
   // DBUF: two ys buffers, ONE barrier per chunk (compute buf[i&1] while
   // draining the next chunk into buf[(i+1)&1])
   unsigned short* ys = reinterpret_cast<unsigned short*>(lds_raw);
   float* topv = reinterpret_cast<float*>(
-      lds_raw + (DBUF ? 2 : 1) * TM_CHUNK * ROWB);
+      lds_raw + (DBUF ? 2 : 1) * CH * ROWB);
   const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
   int* topi = reinterpret_cast<int*>(topv + TM_UPB * KP);
   float* th_lds = reinterpret_cast<float*>(topi + TM_UPB * KP);
@@ -154,7 +154,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   }
 
   // ---- software-pipelined staging registers (granules of 16 B)
-  constexpr int NG = (TM_CHUNK * ROWB) / 16 / 256;  // granules per thread
+  constexpr int NG = (CH * ROWB) / 16 / 256;        // granules per thread
   static_assert(NG >= 1, "chunk must cover one granule per thread");
   u32x4 stg[NG];
   // per-thread element offsets are chunk-invariant: loads use ONE
@@ -170,7 +170,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   }
   auto load_stg = [&](long long cbase) {
     const unsigned short* yb_g = Y + cbase * F;
-    const bool tail = cbase + TM_CHUNK > it1;
+    const bool tail = cbase + CH > it1;
     if (!tail) {
 #pragma unroll
       for (int r = 0; r < NG; ++r)
@@ -194,19 +194,19 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
     }
   };
   auto ybuf = [&](int i) -> unsigned short* {
-    return ys + (DBUF ? (size_t)(i & 1) * TM_CHUNK * F : 0);
+    return ys + (DBUF ? (size_t)(i & 1) * CH * F : 0);
   };
-  const int n_chunks = (int)((it1 - it0 + TM_CHUNK - 1) / TM_CHUNK);
+  const int n_chunks = (int)((it1 - it0 + CH - 1) / CH);
   load_stg(it0);
   if (DBUF) {
     // prologue fill of buffer 0; chunk 1's loads fly under chunk 0
     drain_to(ybuf(0));
     __syncthreads();
-    if (n_chunks > 1) load_stg(it0 + TM_CHUNK);
+    if (n_chunks > 1) load_stg(it0 + CH);
   }
 
   for (int ci = 0; ci < n_chunks; ++ci) {
-    const long long base = it0 + (long long)ci * TM_CHUNK;
+    const long long base = it0 + (long long)ci * CH;
     unsigned short* yb = ybuf(ci);
     if (!DBUF) {
       __syncthreads();  // all waves done reading the previous chunk
@@ -215,7 +215,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
       // issue the NEXT chunk's global loads AFTER the barrier (a
       // __syncthreads compiles to s_waitcnt vmcnt(0)) so they fly
       // under the MFMA phase
-      if (base + TM_CHUNK < it1) load_stg(base + TM_CHUNK);
+      if (base + CH < it1) load_stg(base + CH);
     }
     if (probe) {
       const unsigned long long now = wall_clock64();
@@ -265,8 +265,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
           float th = th_lds[mylist];
           // 32-bit tail guard: li < lim replaces the former 64-bit
           // `item < it1` (16 hoisted v_cmp_gt_i64 in the fast loop)
-          const int lim = (int)(it1 - base < TM_CHUNK ? it1 - base
-                                                      : TM_CHUNK);
+          const int lim = (int)(it1 - base < CH ? it1 - base : CH);
 #pragma unroll
           for (int i = 0; i < IFR; ++i) {
 #pragma unroll
@@ -305,7 +304,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
       // one iteration ago), start chunk ci+2's loads, ONE barrier
       if (ci + 1 < n_chunks) {
         drain_to(ybuf(ci + 1));
-        if (ci + 2 < n_chunks) load_stg(base + 2 * TM_CHUNK);
+        if (ci + 2 < n_chunks) load_stg(base + 2 * CH);
       }
       __syncthreads();
       if (probe) {
@@ -349,9 +348,11 @@ extern "C" void launch_topk_mfma(
   dim3 block(256);
   const char* e_db = getenv("PIO_TOPK_DB");
   const bool use_db = e_db != nullptr && e_db[0] == '1';
+  const char* e_ch = getenv("PIO_TOPK_CHUNK");
+  const int chunk = (e_ch && atoi(e_ch) == 128) ? 128 : TM_CHUNK;
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
-    size_t lds_bytes = (size_t)(use_db ? 2 : 1) * TM_CHUNK * (FF * 2) +      \
+    size_t lds_bytes = (size_t)(use_db ? 2 : 1) * chunk * (FF * 2) +         \
                        (sizeof(float) + sizeof(int)) * TM_UPB * (K + 1) +    \
                        sizeof(float) * TM_UPB;                               \
     static bool attr_set_##FF = false;                                       \
@@ -370,6 +371,11 @@ extern "C" void launch_topk_mfma(
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
                            item_base, prof);                                 \
+      else if (chunk == 128)                                                 \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false, 128>), grid,   \
+                           block, lds_bytes, stream, Xq, Y, item_mask,       \
+                           ban_indptr, ban_indices, out_val, out_idx, B, N,  \
+                           K, n_slices, item_base, prof);                    \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false>), grid, block, \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
@@ -381,6 +387,11 @@ extern "C" void launch_topk_mfma(
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
                            item_base, nullptr);                              \
+      else if (chunk == 128)                                                 \
+        hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false, 128>), grid,  \
+                           block, lds_bytes, stream, Xq, Y, item_mask,       \
+                           ban_indptr, ban_indices, out_val, out_idx, B, N,  \
+                           K, n_slices, item_base, nullptr);                 \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false>), grid,       \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
