@@ -221,3 +221,37 @@ class TestRemoteEndToEnd:
         algo = ALSAlgorithm(ep.algorithms_params[0][1])
         r = algo.predict(models[0], Query(user="u2", num=5))
         assert len(r.item_scores) == 5
+
+
+class TestRemoteConcurrency:
+    def test_concurrent_multi_client_ingest_and_reads(self, remote_storage):
+        """16 client threads hammer one storage daemon with mixed
+        writes/reads — the concurrent multi-process access class the
+        reference gets from PostgreSQL."""
+        import concurrent.futures
+
+        from predictionio_amd.data.storage.base import App
+        apps = remote_storage.get_meta_data_apps()
+        aid = apps.insert(App(0, "ConcApp"))
+        le = remote_storage.get_l_events()
+        le.init(aid)
+
+        def worker(w):
+            n_ok = 0
+            for i in range(10):
+                evs = [mk("view", f"w{w}u{i}_{j}", minutes=w * 100 + i,
+                          target_entity_type="item",
+                          target_entity_id=f"i{j}") for j in range(10)]
+                ids = le.insert_batch(evs, aid)
+                n_ok += len(ids)
+                # interleaved reads
+                got = list(le.find(app_id=aid, entity_id=f"w{w}u{i}_0"))
+                assert len(got) == 1
+            return n_ok
+
+        with concurrent.futures.ThreadPoolExecutor(16) as pool:
+            totals = list(pool.map(worker, range(16)))
+        assert sum(totals) == 16 * 10 * 10
+        cols = le.find_columns(app_id=aid, event_names=["view"])
+        assert len(cols["event"]) == 1600
+        le.remove(aid)
