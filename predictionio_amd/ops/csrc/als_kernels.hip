@@ -1139,11 +1139,7 @@ extern "C" void launch_als_solve(
                        dim3(grid_w2), dim3(128), 0, stream, indptr,          \
                        indices, values, Y, V, X, n_rows, lambda, alpha,      \
                        implicit_mode, wr_scale, prof);                       \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 28, 24, true>),              \
-                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
-                       values, Y, V, X, n_rows, lambda, alpha,               \
-                       implicit_mode, wr_scale, prof);                       \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 28, true>),              \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24, true>),              \
                        dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
                        values, Y, V, X, n_rows, lambda, alpha,               \
                        implicit_mode, wr_scale, prof);                       \
@@ -1172,11 +1168,7 @@ extern "C" void launch_als_solve(
                        dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
                        values, Y, V, X, n_rows, lambda, alpha,               \
                        implicit_mode, wr_scale, prof);                       \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 28, 24, true>),              \
-                       dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
-                       values, Y, V, X, n_rows, lambda, alpha,               \
-                       implicit_mode, wr_scale, prof);                       \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 28, true>),              \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24, true>),              \
                        dim3(grid_w), dim3(128), 0, stream, indptr, indices,  \
                        values, Y, V, X, n_rows, lambda, alpha,               \
                        implicit_mode, wr_scale, prof);                       \
@@ -1189,11 +1181,7 @@ extern "C" void launch_als_solve(
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \
                        X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
                        prof);                                                \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 28, 24>), dim3(grid_w),      \
-                       dim3(128), 0, stream, indptr, indices, values, Y, V,  \
-                       X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
-                       prof);                                                \
-    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 28>), dim3(grid_w),      \
+    hipLaunchKernelGGL((als_woodbury_kernel<FF, 32, 24>), dim3(grid_w),      \
                        dim3(128), 0, stream, indptr, indices, values, Y, V,  \
                        X, n_rows, lambda, alpha, implicit_mode, wr_scale,    \
                        prof);                                                \
