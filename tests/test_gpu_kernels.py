@@ -69,8 +69,8 @@ class TestALSKernel:
         n_cols = 400
         # hits every solver variant boundary: NW=20 (<=20), NW=24
         # (21..24), NW=32 (25..32), dense (>32), empty
-        sizes = [0, 1, 5, 20, 21, 24, 25, 31, 32, 33, 40, 64, 100,
-                 2, 23, 32, 33, 0, 7]
+        sizes = [0, 1, 5, 20, 21, 24, 25, 28, 29, 31, 32, 33, 40, 64,
+                 100, 2, 23, 32, 33, 0, 7]
         rows, cols, vals = [], [], []
         for r, n in enumerate(sizes):
             rows += [r] * n
