@@ -72,8 +72,9 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
-template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK>
-__global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
+template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK,
+          int NWAVES = TM_WAVES>
+__global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
     const uint8_t* __restrict__ item_mask,   // N or nullptr
@@ -95,8 +96,10 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   float* topv = reinterpret_cast<float*>(
       lds_raw + (DBUF ? 2 : 1) * CH * ROWB);
   const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
-  int* topi = reinterpret_cast<int*>(topv + TM_UPB * KP);
-  float* th_lds = reinterpret_cast<float*>(topi + TM_UPB * KP);
+  constexpr int BS = NWAVES * 64;        // block size
+  constexpr int UPB = NWAVES * TM_QPW;   // queries per block
+  int* topi = reinterpret_cast<int*>(topv + UPB * KP);
+  float* th_lds = reinterpret_cast<float*>(topi + UPB * KP);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -108,7 +111,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   unsigned long long pt = 0, acc_setup = 0, acc_stage = 0, acc_score = 0;
   if (probe) pt = wall_clock64();
 
-  const long long u0 = (long long)blockIdx.x * TM_UPB;
+  const long long u0 = (long long)blockIdx.x * UPB;
   const long long guser = u0 + wave * TM_QPW + lq;
   const bool has_user = guser < B;
   const int slice = blockIdx.y;
@@ -130,11 +133,11 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   }
 
   // ---- init the per-query top-K lists + shared thresholds
-  for (int e = tid; e < TM_UPB * KP; e += 256) {
+  for (int e = tid; e < UPB * KP; e += BS) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
-  if (tid < TM_UPB) th_lds[tid] = -FLT_MAX;
+  if (tid < UPB) th_lds[tid] = -FLT_MAX;
   __syncthreads();
   const int mylist = wave * TM_QPW + lq;   // this lane's query list
   float* tvu = topv + mylist * KP;
@@ -154,7 +157,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   }
 
   // ---- software-pipelined staging registers (granules of 16 B)
-  constexpr int NG = (CH * ROWB) / 16 / 256;        // granules per thread
+  constexpr int NG = (CH * ROWB) / 16 / BS;         // granules per thread
   static_assert(NG >= 1, "chunk must cover one granule per thread");
   u32x4 stg[NG];
   // per-thread element offsets are chunk-invariant: loads use ONE
@@ -164,7 +167,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   int grow[NG];
 #pragma unroll
   for (int r = 0; r < NG; ++r) {
-    const int lin = (tid + r * 256) * 16;
+    const int lin = (tid + r * BS) * 16;
     grow[r] = lin / ROWB;
     goff[r] = grow[r] * F + (lin % ROWB) / 2;
   }
@@ -186,7 +189,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   auto drain_to = [&](unsigned short* yb) {
 #pragma unroll
     for (int r = 0; r < NG; ++r) {
-      const int lin = (tid + r * 256) * 16;
+      const int lin = (tid + r * BS) * 16;
       const int row = lin / ROWB;
       const int col = lin % ROWB;
       const int dst = row * ROWB + (col ^ ((row & SWM) << 4));
@@ -318,7 +321,7 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   if (probe) pt = wall_clock64();
 
   // ---- write out: one candidate group per slice per query
-  for (int e = tid; e < TM_UPB * K; e += 256) {
+  for (int e = tid; e < UPB * K; e += BS) {
     const int list = e / K;
     const int q = e % K;
     const long long gu = u0 + list;
@@ -337,6 +340,25 @@ __global__ __launch_bounds__(256, 6) void topk_mfma_kernel(
   }
 }
 
+// 8-wave (128 queries/WG) wide variant: halves the Y stream per query
+// (each staged slice serves 2x the queries). Only meaningful for
+// F >= 64 (the staging granule math needs CH*F*2/16 >= block size).
+template <int FF>
+static void launch_topk_mfma_wide(
+    dim3 grid, size_t lds_bytes, hipStream_t stream,
+    const unsigned short* Xq, const unsigned short* Y,
+    const uint8_t* item_mask, const long long* ban_indptr,
+    const int* ban_indices, float* out_val, int* out_idx,
+    int B, long long N, int K, int n_slices, int item_base) {
+  if constexpr (FF >= 64) {
+    hipLaunchKernelGGL(
+        (topk_mfma_kernel<FF, false, false, TM_CHUNK, 8>), grid,
+        dim3(512), lds_bytes, stream, Xq, Y, item_mask, ban_indptr,
+        ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,
+        nullptr);
+  }
+}
+
 extern "C" void launch_topk_mfma(
     const unsigned short* Xq, const unsigned short* Y,
     const uint8_t* item_mask, const long long* ban_indptr,
@@ -346,10 +368,14 @@ extern "C" void launch_topk_mfma(
 {
   dim3 grid((B + TM_UPB - 1) / TM_UPB, n_slices);
   dim3 block(256);
+  dim3 grid_w((B + 127) / 128, n_slices);
   const char* e_db = getenv("PIO_TOPK_DB");
   const bool use_db = e_db != nullptr && e_db[0] == '1';
   const char* e_ch = getenv("PIO_TOPK_CHUNK");
   const int chunk = (e_ch && atoi(e_ch) == 128) ? 128 : TM_CHUNK;
+  const char* e_w = getenv("PIO_TOPK_WIDE");
+  const bool use_wide = e_w != nullptr && e_w[0] == '1' && f >= 64 &&
+                        prof == nullptr && !use_db && chunk == TM_CHUNK;
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
     size_t lds_bytes = (size_t)(use_db ? 2 : 1) * chunk * (FF * 2) +         \
@@ -381,6 +407,13 @@ extern "C" void launch_topk_mfma(
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
                            item_base, prof);                                 \
+    } else if (use_wide) {                                                   \
+      size_t lds_w = (size_t)TM_CHUNK * (FF * 2) +                           \
+                     (sizeof(float) + sizeof(int)) * 128 * (K + 1) +         \
+                     sizeof(float) * 128;                                    \
+      launch_topk_mfma_wide<FF>(grid_w, lds_w, stream, Xq, Y, item_mask,     \
+                                ban_indptr, ban_indices, out_val, out_idx,   \
+                                B, N, K, n_slices, item_base);               \
     } else {                                                                 \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, true>), grid, block, \
