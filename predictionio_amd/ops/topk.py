@@ -66,6 +66,12 @@ def topk_score(Xq: torch.Tensor, Y: torch.Tensor, K: int,
     N = Y.shape[0]
     K = int(K)
     if Xq.is_cuda:
+        if K > 64:
+            # the fused kernels keep K-entry per-query lists in LDS and
+            # cap at 64; arbitrary `num` (the reference allows any) falls
+            # back to a materialized masked matmul + torch.topk on device
+            return _topk_torch_gpu(Xq, Y, K, item_mask, ban_indptr,
+                                   ban_indices)
         if mode is None:
             mode = os.environ.get("PIO_TOPK_MODE", "mfma")
         if mode == "mfma" and _mfma_rank(f) is not None:
@@ -97,6 +103,36 @@ def topk_score(Xq: torch.Tensor, Y: torch.Tensor, K: int,
         mvals = mvals.masked_fill(empty, float("-inf"))
         return mvals, midx
     return topk_score_ref(Xq, Y, K, item_mask, ban_indptr, ban_indices)
+
+
+def _topk_torch_gpu(Xq: torch.Tensor, Y: torch.Tensor, K: int,
+                    item_mask, ban_indptr, ban_indices
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Device fallback for K > 64: materialized masked score matrix +
+    torch.topk. Memory: B x N fp32 — callers with huge B should batch."""
+    scores = Xq @ Y.t()
+    if item_mask is not None:
+        scores = scores.masked_fill(item_mask.bool().unsqueeze(0),
+                                    float("-inf"))
+    if ban_indptr is not None:
+        ip = ban_indptr.tolist()
+        for b in range(scores.shape[0]):
+            banned = ban_indices[ip[b]:ip[b + 1]].long()
+            if banned.numel():
+                scores[b, banned] = float("-inf")
+    Kc = min(K, scores.shape[1])
+    vals, idxs = torch.topk(scores, Kc, dim=1)
+    idxs = idxs.clone()
+    idxs[vals == float("-inf")] = -1
+    if Kc < K:
+        B = scores.shape[0]
+        vals = torch.cat(
+            [vals, torch.full((B, K - Kc), float("-inf"),
+                              device=vals.device)], 1)
+        idxs = torch.cat(
+            [idxs, torch.full((B, K - Kc), -1, dtype=idxs.dtype,
+                              device=idxs.device)], 1)
+    return vals, idxs
 
 
 def _topk_score_mfma(Xq: torch.Tensor, Y: torch.Tensor, K: int,

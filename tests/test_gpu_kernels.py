@@ -442,3 +442,23 @@ class TestRCCLBackend:
             cwd=repo)
         assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
         assert "RCCL smoke OK" in r.stdout
+
+
+@requires_gpu
+class TestLargeK:
+    def test_k_above_kernel_cap_falls_back(self):
+        """The reference allows arbitrary `num`; K > 64 takes the
+        device torch fallback and must match the CPU reference."""
+        from predictionio_amd.ops import topk as topk_ops
+        g = torch.Generator().manual_seed(71)
+        B, N, f, K = 9, 3000, 64, 100
+        Xq = torch.randn((B, f), generator=g).float()
+        Y = torch.randn((N, f), generator=g).float()
+        mask = (torch.rand(N, generator=g) < 0.2).to(torch.uint8)
+        rv, ri = topk_ops.topk_score_ref(Xq, Y, K, item_mask=mask)
+        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K,
+                                     item_mask=mask.cuda())
+        assert torch.allclose(gv.cpu(), rv, atol=1e-4, rtol=1e-4)
+        chosen = (Xq @ Y.t()).gather(1, gi.cpu().clamp_min(0))
+        ok = rv != float("-inf")
+        assert torch.allclose(chosen[ok], rv[ok], atol=1e-4, rtol=1e-4)
