@@ -101,7 +101,8 @@ def _get_client(source_name: str, cfg: dict):
             from predictionio_amd.data.storage.remote import RemoteClient
             client = RemoteClient(
                 cfg.get("url", "http://127.0.0.1:7072"),
-                timeout=float(cfg.get("timeout", "30")))
+                timeout=float(cfg.get("timeout", "30")),
+                key=cfg.get("key"))
         else:
             raise StorageError(f"Unknown storage source type: {typ}")
         _sources[key] = client

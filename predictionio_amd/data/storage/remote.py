@@ -121,10 +121,13 @@ def opt_unfield(d: Optional[Dict[str, Any]]):
 class RemoteClient:
     """One storage source = one storage-server URL."""
 
-    def __init__(self, url: str, timeout: float = 30.0):
+    def __init__(self, url: str, timeout: float = 30.0,
+                 key: str = None):
         import httpx
         self.url = url.rstrip("/")
-        self._http = httpx.Client(base_url=self.url, timeout=timeout)
+        headers = {"X-PIO-Storage-Key": key} if key else None
+        self._http = httpx.Client(base_url=self.url, timeout=timeout,
+                                  headers=headers)
 
     def call(self, kind: str, method: str, payload: Dict[str, Any]) -> Any:
         r = self._http.post(f"/s/{kind}/{method}", json=payload)

@@ -228,19 +228,32 @@ def create_app(sqlite_path: str = None) -> FastAPI:
                         for k, v in out.items()}
         raise ValueError(f"unknown rpc {kind}.{method}")
 
+    # optional shared-secret auth (the PostgreSQL-password analog):
+    # set PIO_STORAGE_SERVER_KEY on the daemon; clients send it via the
+    # X-PIO-Storage-Key header (remote driver: source config KEY=...)
+    import os as _os
+    auth_key = _os.environ.get("PIO_STORAGE_SERVER_KEY")
+
     @app.get("/")
     def index():
         return {"status": "alive", "service": "pio-storage-server"}
 
     @app.post("/s/{kind}/{method}")
     async def rpc(kind: str, method: str, request: Request):
+        if auth_key and request.headers.get("X-PIO-Storage-Key") != auth_key:
+            return JSONResponse({"message": "invalid storage key"},
+                                status_code=401)
         try:
             payload = await request.json()
         except Exception:
             return JSONResponse({"message": "invalid JSON"},
                                 status_code=400)
         try:
-            return {"r": handle(kind, method, payload)}
+            # DB work runs in the threadpool so slow queries don't
+            # stall the event loop for other clients
+            from starlette.concurrency import run_in_threadpool
+            return {"r": await run_in_threadpool(handle, kind, method,
+                                                 payload)}
         except ValueError as e:
             return JSONResponse({"message": str(e)}, status_code=404)
         except Exception as e:  # noqa: BLE001
