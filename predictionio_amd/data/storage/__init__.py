@@ -10,8 +10,11 @@ source type at runtime, like the reference's reflective
 (Storage.scala:310-359).
 
 Defaults (no env set): a single SQLITE source at $PIO_FS_BASEDIR/pio.sqlite
-serving all three repositories. Backends available: sqlite (all three),
-localfs (modeldata), memory (sqlite :memory:, tests).
+serving all three repositories. Backends available: sqlite (all three,
+embedded), remote (all three — client of the `pio storageserver` daemon;
+keys: URL, [TIMEOUT], [KEY] matching the daemon's
+PIO_STORAGE_SERVER_KEY), localfs + fsspec (modeldata), memory
+(sqlite :memory:, tests).
 """
 
 from __future__ import annotations
