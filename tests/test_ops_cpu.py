@@ -214,3 +214,35 @@ class TestPypio:
         cols = pypio.events_to_columns(evs)
         assert cols["entityId"] == ["u1"]
         assert cols["properties"][0]["rating"] == 5
+
+
+class TestBf16Cache:
+    def test_cache_hits_and_invalidation(self):
+        import torch
+
+        from predictionio_amd.ops.topk import bf16_copy
+        Y = torch.randn(8, 4)
+        a = bf16_copy(Y)
+        assert a.dtype == torch.bfloat16 and a.shape == (8, 4)
+        assert bf16_copy(Y) is a          # cache hit
+        Y[0, 0] = 42.0                    # in-place write bumps _version
+        b = bf16_copy(Y)
+        assert b is not a
+        assert float(b[0, 0]) == 42.0
+        # padded variant
+        c = bf16_copy(Y, pad_to=8)
+        assert c.shape == (8, 8) and float(c[0, 7]) == 0.0
+
+    def test_cache_releases_tensor(self):
+        import gc
+        import weakref
+
+        import torch
+
+        from predictionio_amd.ops.topk import bf16_copy
+        Y = torch.randn(4, 4)
+        bf16_copy(Y)
+        r = weakref.ref(Y)
+        del Y
+        gc.collect()
+        assert r() is None  # weak keying: no leak of the fp32 factors
