@@ -317,3 +317,19 @@ class TestFindColumns:
         with pytest.raises(ValueError):
             events.find_columns(app_id=1,
                                 property_fields=["x'); DROP TABLE t;--"])
+
+    def test_channel_scoped_columns(self, mem_storage):
+        le = mem_storage.get_l_events()
+        le.init(2)
+        le.init(2, 5)  # channel 5
+        le.insert(mk("rate", "u1", 0, target_entity_type="item",
+                     target_entity_id="i1",
+                     properties=DataMap({"rating": 1.0})), 2)
+        le.insert(mk("rate", "u2", 0, target_entity_type="item",
+                     target_entity_id="i2",
+                     properties=DataMap({"rating": 2.0})), 2, 5)
+        default = le.find_columns(app_id=2, property_fields=["rating"])
+        chan = le.find_columns(app_id=2, channel_id=5,
+                               property_fields=["rating"])
+        assert default["entity_id"] == ["u1"]
+        assert chan["entity_id"] == ["u2"] and chan["rating"] == [2.0]
