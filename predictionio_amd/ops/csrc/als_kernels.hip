@@ -360,7 +360,12 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
-  constexpr int PD = 4;  // global-load pipeline depth (latency hiding)
+  constexpr int PD = 4;  // global-load pipeline depth (latency hiding).
+  // PD=8 and PD=12 MEASURED SLOWER (round 2: nnz=200 23.2->23.9/26.8 ms
+  // per 0.2M rows) despite LDS capping occupancy either way — the extra
+  // in-flight registers hurt scheduling more than the added latency
+  // coverage helps; the per-item LDS broadcast + two wave_syncs, not
+  // HBM latency, bound this loop
   constexpr int LP = F + 4;  // Lc row stride: 16B-aligned for b128 reads
   __shared__ float ys[2][PD][F];       // staged y slots per wave
   __shared__ float Lc[2][F][LP];       // persisted L columns: Lc[w][k][j] = L[j][k]
