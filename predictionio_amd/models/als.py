@@ -39,6 +39,11 @@ class ALSParams:
     gather_dtype: Optional[str] = None  # "bf16" halves multi-GPU factor
                                         # gather bytes; numerics study in
                                         # profiles/bf16_numerics_study.txt
+    checkpoint_every: int = 0   # >0: persist local factor shards every N
+                                # iterations (the reference checkpoints
+    checkpoint_dir: Optional[str] = None  # ALS every 10 iters to bound
+                                # lineage, ALSAlgorithm.scala:85; here it
+                                # is crash-resume for 100M-scale runs)
 
 
 class ALSTrainer:
@@ -193,16 +198,70 @@ class ALSTrainer:
             self.phase_times["gather_s"] += (t1 - t0) + (t3 - t2)
             self.phase_times["solve_s"] += (t2 - t1) + (t4 - t3)
 
+    # ------------------------------------------------------- checkpointing
+
+    def _ckpt_path(self) -> Optional[str]:
+        if not self.p.checkpoint_every or not self.p.checkpoint_dir:
+            return None
+        os.makedirs(self.p.checkpoint_dir, exist_ok=True)
+        return os.path.join(self.p.checkpoint_dir,
+                            f"als_ckpt_rank{pdist.get_rank()}.pt")
+
+    def save_checkpoint(self, iteration: int) -> None:
+        """Persist this rank's LOCAL factor shards + iteration counter
+        (atomic rename so a crash mid-write keeps the previous one)."""
+        path = self._ckpt_path()
+        if path is None:
+            return
+        tmp = path + ".tmp"
+        torch.save({"iteration": iteration,
+                    "X": self.X.detach().cpu(),
+                    "Y": self.Y.detach().cpu(),
+                    "rank": self.p.rank,
+                    "u_block": (self.u_lo, self.u_hi),
+                    "i_block": (self.i_lo, self.i_hi)}, tmp)
+        os.replace(tmp, path)
+
+    def load_checkpoint(self) -> int:
+        """Resume local shards if a matching checkpoint exists; returns
+        the iteration to continue FROM (0 = fresh start)."""
+        path = self._ckpt_path()
+        if path is None or not os.path.exists(path):
+            return 0
+        blob = torch.load(path, weights_only=True)
+        if (blob.get("rank") != self.p.rank
+                or tuple(blob.get("u_block", ())) != (self.u_lo, self.u_hi)
+                or tuple(blob.get("i_block", ())) != (self.i_lo, self.i_hi)):
+            return 0  # shape/shard mismatch: ignore stale checkpoint
+        self.X = blob["X"].to(self.device)
+        self.Y = blob["Y"].to(self.device)
+        self._pending_y = None
+        return int(blob["iteration"])
+
     def fit(self) -> Tuple[torch.Tensor, torch.Tensor]:
         """Train and return the FULL (n_users x f, n_items x f) factor
         matrices. Under multi-GPU training each rank only solves its row
         blocks, so the local shards are all-gathered at the end — the
         persisted model must cover the whole catalog (every template saves
-        fit()'s return value against the full user_map/item_map)."""
+        fit()'s return value against the full user_map/item_map).
+
+        With checkpoint_every/checkpoint_dir set, local shards are
+        persisted every N iterations and fit() resumes from the latest
+        checkpoint after a crash (every rank must see the same
+        checkpoint_dir state)."""
+        start = self.load_checkpoint() if self._ckpt_path() else 0
         if self.X is None:
             self.init_factors()
-        for _ in range(self.p.iterations):
+        for it in range(start, self.p.iterations):
             self.step()
+            if (self.p.checkpoint_every
+                    and (it + 1) % self.p.checkpoint_every == 0
+                    and it + 1 < self.p.iterations):
+                if self._pending_y is not None:
+                    # materialize the in-flight gather so local Y is final
+                    self._pending_y.finish()
+                    self._pending_y = None
+                self.save_checkpoint(it + 1)
         if pdist.is_distributed():
             return self.gather_factors()
         return self.X, self.Y

@@ -274,7 +274,11 @@ class ALSAlgorithm(Algorithm):
             lambda_=float(self.params.get("lambda", 0.01)),
             alpha=float(self.params.get("alpha", 1.0)),
             implicit=implicit,
-            seed=self.params.get("seed"))
+            seed=self.params.get("seed"),
+            # reference checkpoints ALS every 10 iterations
+            # (ALSAlgorithm.scala:85); here it is crash-resume
+            checkpoint_every=int(self.params.get("checkpointEvery", 0)),
+            checkpoint_dir=self.params.get("checkpointDir"))
         device = torch.device("cuda") if torch.cuda.is_available() \
             else torch.device("cpu")
         X, Y = train_als(users, items, vals, len(user_map), len(item_map),

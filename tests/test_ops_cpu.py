@@ -246,3 +246,60 @@ class TestBf16Cache:
         del Y
         gc.collect()
         assert r() is None  # weak keying: no leak of the fp32 factors
+
+
+class TestALSCheckpoint:
+    def test_resume_matches_uninterrupted(self, tmp_path):
+        import torch
+
+        from predictionio_amd.models.als import ALSParams, ALSTrainer
+        g = torch.Generator().manual_seed(21)
+        n_u, n_i, f = 30, 20, 16
+        users = torch.randint(0, n_u, (200,), generator=g,
+                              dtype=torch.int32)
+        items = torch.randint(0, n_i, (200,), generator=g,
+                              dtype=torch.int32)
+        vals = (torch.rand(200, generator=g) * 4 + 1).float()
+
+        def mk(ckpt):
+            p = ALSParams(rank=f, iterations=4, lambda_=0.05, seed=7,
+                          checkpoint_every=2 if ckpt else 0,
+                          checkpoint_dir=str(tmp_path) if ckpt else None)
+            t = ALSTrainer(p, n_u, n_i, torch.device("cpu"))
+            t.set_ratings(users, items, vals)
+            return t
+
+        ref = mk(False)
+        Xr, Yr = ref.fit()
+
+        # run WITH checkpointing: a ckpt lands after iteration 2
+        t1 = mk(True)
+        X1, Y1 = t1.fit()
+        assert torch.allclose(X1, Xr) and torch.allclose(Y1, Yr)
+        import os
+        assert any(fn.startswith("als_ckpt") for fn in os.listdir(tmp_path))
+
+        # "crash" and resume: a fresh trainer picks up at iteration 2
+        t2 = mk(True)
+        assert t2.load_checkpoint() == 2
+        X2, Y2 = t2.fit()  # fit() itself resumes from the checkpoint
+        assert torch.allclose(X2, Xr, atol=1e-6)
+        assert torch.allclose(Y2, Yr, atol=1e-6)
+
+    def test_stale_checkpoint_ignored(self, tmp_path):
+        import torch
+
+        from predictionio_amd.models.als import ALSParams, ALSTrainer
+        p = ALSParams(rank=16, iterations=1, checkpoint_every=1,
+                      checkpoint_dir=str(tmp_path), seed=1)
+        t = ALSTrainer(p, 10, 8, torch.device("cpu"))
+        t.set_ratings(torch.tensor([0, 1], dtype=torch.int32),
+                      torch.tensor([0, 1], dtype=torch.int32),
+                      torch.ones(2))
+        t.fit()
+        t.save_checkpoint(1)
+        # different shard shape -> checkpoint must be ignored
+        p2 = ALSParams(rank=16, iterations=1, checkpoint_every=1,
+                       checkpoint_dir=str(tmp_path), seed=1)
+        t2 = ALSTrainer(p2, 12, 8, torch.device("cpu"))
+        assert t2.load_checkpoint() == 0
