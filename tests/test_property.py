@@ -214,3 +214,51 @@ class TestFindFilterProperties:
         exp_keys = sorted((e.event, e.entity_id,
                            e.event_time.isoformat()) for e in expect)
         assert got_keys == exp_keys
+
+
+class TestRemoteWireRoundtrip:
+    """Property tests for the remote-backend wire marshalling."""
+
+    @given(st.one_of(st.none(), st.text(max_size=20)))
+    def test_opt_field_tristate(self, v):
+        from predictionio_amd.data.storage.remote import (
+            opt_field, opt_unfield,
+        )
+        from predictionio_amd.data.storage.base import UNSET
+        assert opt_unfield(opt_field(v)) == v
+        assert opt_unfield(opt_field(UNSET)) is UNSET
+
+    @given(st.datetimes(min_value=datetime(1980, 1, 1),
+                        max_value=datetime(2200, 1, 1)))
+    def test_ms_roundtrip(self, dt):
+        from datetime import timezone
+
+        from predictionio_amd.data.storage.remote import from_ms, ms
+        aware = dt.replace(tzinfo=timezone.utc)
+        back = from_ms(ms(aware))
+        assert abs((back - aware).total_seconds()) < 0.001
+
+    @given(st.text(min_size=1, max_size=30),
+           st.text(min_size=1, max_size=30),
+           st.dictionaries(st.text(min_size=1, max_size=8).filter(
+               lambda s: not s.startswith("$") and not s.startswith("pio_")),
+               st.one_of(st.integers(-10**6, 10**6), st.text(max_size=12),
+                         st.booleans()), max_size=4))
+    def test_engine_instance_roundtrip(self, fac, var, env):
+        from predictionio_amd.data.storage.base import (
+            EngineInstance, utcnow,
+        )
+        from predictionio_amd.data.storage.remote import (
+            ei_from_dict, ei_to_dict,
+        )
+        env = {k: str(v) for k, v in env.items()}
+        now = utcnow().replace(microsecond=0)
+        i = EngineInstance(
+            id="x1", status="COMPLETED", start_time=now, end_time=now,
+            engine_id="e", engine_version="1", engine_variant=var,
+            engine_factory=fac, batch="b", env=env,
+            data_source_params='{"a": 1}')
+        j = ei_from_dict(ei_to_dict(i))
+        assert j.engine_factory == fac and j.engine_variant == var
+        assert j.env == env and j.data_source_params == '{"a": 1}'
+        assert j.start_time == now
