@@ -73,7 +73,7 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
 }
 
 template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK,
-          int NWAVES = TM_WAVES>
+          int NWAVES = TM_WAVES, bool GLL = false>
 __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
@@ -199,8 +199,32 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
   auto ybuf = [&](int i) -> unsigned short* {
     return ys + (DBUF ? (size_t)(i & 1) * CH * F : 0);
   };
+  // GLL: async global->LDS DMA (no staging registers, no drain
+  // ds_writes). The LDS destination is wave-uniform base + lane*16
+  // (linear), so the XOR swizzle moves to the per-lane GLOBAL source
+  // address — the involution makes the LDS image identical to a
+  // swizzled write (guide rule 21). Tail rows clamp the source row;
+  // their stale granules are filtered by the epilogue's li < lim.
+  auto issue_gll = [&](long long cbase) {
+#pragma unroll
+    for (int r = 0; r < NG; ++r) {
+      const int g = tid + r * BS;
+      const int lin = g * 16;
+      const int row = lin / ROWB;
+      const int col = lin % ROWB;
+      long long grow_g = cbase + row;
+      if (grow_g >= N) grow_g = N - 1;
+      const unsigned short* src =
+          Y + grow_g * F + ((col ^ ((row & SWM) << 4)) >> 1);
+      unsigned short* dst =
+          ys + ((size_t)(tid & ~63) + (size_t)r * BS) * 8;  // wave base
+      __builtin_amdgcn_global_load_lds(
+          reinterpret_cast<const unsigned int*>(src),
+          reinterpret_cast<unsigned int*>(dst), 16, 0, 0);
+    }
+  };
   const int n_chunks = (int)((it1 - it0 + CH - 1) / CH);
-  load_stg(it0);
+  if (!GLL) load_stg(it0);
   if (DBUF) {
     // prologue fill of buffer 0; chunk 1's loads fly under chunk 0
     drain_to(ybuf(0));
@@ -211,7 +235,11 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
   for (int ci = 0; ci < n_chunks; ++ci) {
     const long long base = it0 + (long long)ci * CH;
     unsigned short* yb = ybuf(ci);
-    if (!DBUF) {
+    if (GLL) {
+      __syncthreads();  // all waves done reading the previous chunk
+      issue_gll(base);
+      __syncthreads();  // drains the in-flight LDS DMA (vmcnt)
+    } else if (!DBUF) {
       __syncthreads();  // all waves done reading the previous chunk
       drain_to(yb);
       __syncthreads();
@@ -373,6 +401,8 @@ extern "C" void launch_topk_mfma(
   const bool use_db = e_db != nullptr && e_db[0] == '1';
   const char* e_ch = getenv("PIO_TOPK_CHUNK");
   const int chunk = (e_ch && atoi(e_ch) == 128) ? 128 : TM_CHUNK;
+  const char* e_g = getenv("PIO_TOPK_GLL");
+  const bool use_gll = e_g != nullptr && e_g[0] == '1';
   const char* e_w = getenv("PIO_TOPK_WIDE");
   const bool use_wide = e_w != nullptr && e_w[0] == '1' && f >= 64 &&
                         prof == nullptr && !use_db && chunk == TM_CHUNK;
@@ -425,6 +455,12 @@ extern "C" void launch_topk_mfma(
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
                            K, n_slices, item_base, nullptr);                 \
+      else if (use_gll)                                                      \
+        hipLaunchKernelGGL(                                                  \
+            (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, true>),  \
+            grid, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
+            ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
+            nullptr);                                                        \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false>), grid,       \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
