@@ -286,10 +286,18 @@ def create_app(config: ServerConfig,
                     batcher.start()
                 prediction = await batcher.submit(query)
             else:
-                supplemented = s.serving.supplement(query)
-                predictions = [a.predict(m, supplemented)
-                               for a, m in zip(s.algorithms, s.models)]
-                prediction = s.serving.serve(query, predictions)
+                # threadpool: predict may do model compute AND live
+                # event-store lookups (ecommerce's 200 ms budget) —
+                # neither may stall the event loop for other requests
+                from starlette.concurrency import run_in_threadpool
+
+                def _predict_one():
+                    supplemented = s.serving.supplement(query)
+                    predictions = [a.predict(m, supplemented)
+                                   for a, m in zip(s.algorithms, s.models)]
+                    return s.serving.serve(query, predictions)
+
+                prediction = await run_in_threadpool(_predict_one)
         except Exception as e:
             logger.exception("query failed")
             if config.log_url:

@@ -63,8 +63,12 @@ def run_train(engine: Engine,
     collectives require all ranks — but only rank 0 touches storage."""
     from predictionio_amd.parallel import dist as pdist
     if pdist.get_rank() != 0:
-        engine.train(engine_params, skip_sanity_check=skip_sanity_check)
-        return ""
+        try:
+            engine.train(engine_params,
+                         skip_sanity_check=skip_sanity_check)
+            return ""
+        finally:
+            run_cleanup()
     instances = storage.get_meta_data_engine_instances()
     instance = EngineInstance(
         id="", status="INIT", start_time=utcnow(), end_time=utcnow(),
