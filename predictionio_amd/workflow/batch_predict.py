@@ -29,9 +29,13 @@ def run_batch_predict(variant: Dict, input_path: str, output_path: str,
     with open(input_path) as f:
         queries = [json.loads(line) for line in f if line.strip()]
 
+    # typed queries when the algorithm declares a parser (the server
+    # route does the same; JsonExtractor semantics)
+    conv = getattr(st.algorithms[0], "query_from_json", None)
+    typed = [conv(q) if conv else q for q in queries]
     # supplement each query, then let each algorithm batch-predict (device
     # algorithms fuse this into one kernel launch)
-    supplemented = [st.serving.supplement(q) for q in queries]
+    supplemented = [st.serving.supplement(q) for q in typed]
     indexed = list(enumerate(supplemented))
     per_algo = []
     for algo, model in zip(st.algorithms, st.models):
@@ -41,7 +45,7 @@ def run_batch_predict(variant: Dict, input_path: str, output_path: str,
     with open(output_path, "w") as out:
         for i, q in enumerate(queries):
             preds = [pa[i] for pa in per_algo]
-            result = st.serving.serve(q, preds)
+            result = st.serving.serve(typed[i], preds)
             rj = result.to_json() if hasattr(result, "to_json") else result
             out.write(json.dumps({"query": q, "prediction": rj}) + "\n")
             n += 1
