@@ -22,6 +22,19 @@ import torch
 
 from predictionio_amd.ops import topk as topk_ops
 
+# all GraphedTopK captures share ONE graph mempool: a second capture
+# while an earlier graph's PRIVATE pool is live (with eager work
+# interleaved) memory-faulted on ROCm (round-2 repro) — the shared-pool
+# pattern is the documented multi-capture arrangement
+_shared_pool = None
+
+
+def _graph_pool():
+    global _shared_pool
+    if _shared_pool is None:
+        _shared_pool = torch.cuda.graph_pool_handle()
+    return _shared_pool
+
 
 class GraphedTopK:
     def __init__(self, Y: torch.Tensor, K: int, batch: int,
@@ -39,14 +52,16 @@ class GraphedTopK:
                   ban_indices=ban_indices)
         # warm up on a side stream (allocator state must be stable before
         # capture)
+        torch.cuda.synchronize()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(warmup):
                 topk_ops.topk_score(self._xq, Y, K, **kw)
         torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        with torch.cuda.graph(self._graph, pool=_graph_pool()):
             self._out_v, self._out_i = topk_ops.topk_score(
                 self._xq, Y, K, **kw)
 
