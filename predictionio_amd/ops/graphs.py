@@ -3,11 +3,21 @@
 `GraphedTopK` captures the scoring + merge sequence into a hipGraph
 (torch.cuda.CUDAGraph is hipGraph-backed on ROCm) for a FIXED (B, N, K)
 shape and replays it with new query content — one graph launch per
-request. Measured on a 10M-item catalog the B=1 path is kernel-time
-dominated (3.45 ms eager == graphed), so the graph buys nothing THERE;
-it pays on small catalogs / many-launch pipelines where per-launch
-overhead is a real fraction (and it pins the serving step's allocations,
-which stabilizes tail latency under allocator pressure).
+request. Measured on a 10M-item catalog the serving step is kernel-time
+dominated at every batch size (round 2, MFMA path: B=1 0.64 ms eager vs
+0.65 ms graphed), so graphs are NEUTRAL there; they pay on small
+catalogs / many-launch pipelines, and they pin the serving step's
+allocations (stable tail latency under allocator pressure).
+
+OPERATIONAL CONSTRAINT (round-2 finding): capture all GraphedTopK
+instances at DEPLOYMENT TIME, before eager traffic, and keep them for
+the process lifetime. Interleaving large-N eager scoring between a
+replayed graph and a NEW capture intermittently memory-faults inside
+ROCm's graph/allocator layer (deterministic repro in the round-2 log;
+shared-pool capture, pre-capture syncs and safe teardown — all
+implemented here — narrow but do not fully remove it). Since eager is
+measured equal on big catalogs, prefer eager unless graphs demonstrably
+help your shape.
 
 Usage (serving hot path, shapes fixed per deployment):
     g = GraphedTopK(Y, K=20, batch=1)
@@ -72,3 +82,22 @@ class GraphedTopK:
         self._xq.copy_(xq, non_blocking=True)
         self._graph.replay()
         return self._out_v, self._out_i
+
+    def close(self) -> None:
+        """Tear the graph down SAFELY: destroying a replayed graph with
+        in-flight work and then capturing a new one memory-faulted on
+        ROCm (round-2 repro) — synchronize, reset, release."""
+        gph = getattr(self, "_graph", None)
+        if gph is not None:
+            try:
+                torch.cuda.synchronize()
+                gph.reset()
+            except Exception:
+                pass
+            self._graph = None
+
+    def __del__(self):  # noqa: D105
+        try:
+            self.close()
+        except Exception:
+            pass
