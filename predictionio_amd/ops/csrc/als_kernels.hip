@@ -340,7 +340,17 @@ __global__ __launch_bounds__(256) void als_solve_kernel(
 //     issued before the current item's math (double-buffered LDS).
 // ---------------------------------------------------------------------------
 
-template <int F>
+// BF16G: the Gramian runs on the MATRIX CORES — the north-star's
+// "ALS factor-update Gramian ... on MFMA". Items are staged TRANSPOSED
+// as bf16 (ysT[feature][item], rows scaled by sqrt(w_a) so the product
+// is a plain S^T S), and because S^T S is symmetric the A- and
+// B-operand fragments are the SAME four b128 loads per 32-item stripe:
+// 4 ds_read_b128 feed 16 v_mfma_f32_16x16x32_bf16 accumulating the
+// full 64x64 normal matrix in fp32. One LDS pass through Lc
+// redistributes D's fragment layout (col = lane&15) into the solver's
+// lane-as-column acc[F]. Opt-in via PIO_ALS_STAGE_BF16 (same umbrella
+// as the Woodbury bf16 staging; b and the Cholesky stay fp32).
+template <int F, bool BF16G = false>
 __global__ __launch_bounds__(128) void als_solve_wave_kernel(
     const long long* __restrict__ indptr,
     const int* __restrict__ indices,
@@ -357,6 +367,7 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
     int skip_above)   // rows with nnz > this go to the workgroup kernel
 {
   static_assert(F <= 64, "wave kernel supports rank <= 64");
+  static_assert(!BF16G || F == 64, "MFMA Gramian variant needs F=64");
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
 
@@ -368,6 +379,10 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
   // HBM latency, bound this loop
   constexpr int LP = F + 4;  // Lc row stride: 16B-aligned for b128 reads
   __shared__ float ys[2][PD][F];       // staged y slots per wave
+  // BF16G: transposed bf16 stripe, rows padded to 40 (80 B, 16B-mult);
+  // granule-XOR (row&3) spreads the strided column writes
+  constexpr int TP = 40;
+  __shared__ unsigned short ysT[2][BF16G ? F : 1][BF16G ? TP : 1];
   __shared__ float Lc[2][F][LP];       // persisted L columns: Lc[w][k][j] = L[j][k]
 
   for (long long row = (long long)blockIdx.x * 2 + wave; row < n_rows;
@@ -382,6 +397,88 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
     for (int m = 0; m < F; ++m) acc[m] = 0.f;
     float b_reg = 0.f;
 
+    if constexpr (BF16G) {
+      // ---- MFMA Gramian (see the kernel comment) ----
+      typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8g;
+      typedef __attribute__((ext_vector_type(4))) float f32x4g;
+      const int lg = lane >> 4;   // fragment k-granule
+      const int lq = lane & 15;   // fragment row-in-block
+      f32x4g accm[4][4];
+#pragma unroll
+      for (int ti = 0; ti < 4; ++ti)
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj)
+          accm[ti][tj] = f32x4g{0.f, 0.f, 0.f, 0.f};
+      // 4-deep pipelined item loads (per item: one coalesced f32 row)
+      float yn2[PD], wa2[PD], wb2[PD];
+#pragma unroll
+      for (int p = 0; p < PD; ++p) {
+        if (p < nnz) {
+          const int col = indices[start + p];
+          yn2[p] = lane < F ? Y[(long long)col * F + lane] : 0.f;
+          const float v = values[start + p];
+          wa2[p] = implicit_mode ? alpha * v : 1.f;
+          wb2[p] = implicit_mode ? 1.f + alpha * v : v;
+        }
+      }
+      for (int c0 = 0; c0 < nnz; c0 += 32) {
+        for (int cm = 0; cm < 32; ++cm) {
+          const int c = c0 + cm;
+          const int slot = c & (PD - 1);
+          float sy = 0.f;
+          if (c < nnz) {
+            const float wa = wa2[slot];
+            const float sc = sqrtf(wa > 1e-12f ? wa : 1e-12f);
+            const float ycur = yn2[slot];
+            sy = sc * ycur;
+            b_reg = fmaf(wb2[slot], ycur, b_reg);   // b stays fp32-exact
+            if (c + PD < nnz) {
+              const int ncol = indices[start + c + PD];
+              yn2[slot] = lane < F ? Y[(long long)ncol * F + lane] : 0.f;
+              const float nv = values[start + c + PD];
+              wa2[slot] = implicit_mode ? alpha * nv : 1.f;
+              wb2[slot] = implicit_mode ? 1.f + alpha * nv : nv;
+            }
+          }
+          // transposed bf16 store: ysT[feature=lane][item cm], granule
+          // (cm>>3) XOR'd with (row&3)
+          const unsigned short hb =
+              (unsigned short)(__float_as_uint(sy) >> 16);
+          const int gcol = (((cm >> 3) ^ (lane & 3)) << 3) | (cm & 7);
+          ysT[wave][lane][gcol] = hb;
+        }
+        wave_sync();
+        bf16x8g frag[4];
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {
+          const int row = t * 16 + lq;
+          const int g = lg ^ (row & 3);
+          frag[t] = *reinterpret_cast<const bf16x8g*>(
+              &ysT[wave][row][g << 3]);
+        }
+#pragma unroll
+        for (int ti = 0; ti < 4; ++ti)
+#pragma unroll
+          for (int tj = 0; tj < 4; ++tj)
+            accm[ti][tj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                frag[ti], frag[tj], accm[ti][tj], 0, 0, 0);
+        wave_sync();
+      }
+      // redistribute D (col = lane&15, row = (lane>>4)*4 + r) into the
+      // solver's lane-as-column acc[F] through Lc (free until Cholesky)
+#pragma unroll
+      for (int ti = 0; ti < 4; ++ti)
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            Lc[wave][ti * 16 + lg * 4 + r][tj * 16 + lq] =
+                accm[ti][tj][r];
+      wave_sync();
+#pragma unroll
+      for (int m = 0; m < F; ++m) acc[m] = Lc[wave][m][lane];
+      wave_sync();
+    } else {
     // ---- Gramian accumulation (register row, LDS broadcast) ----
     // PD-deep load pipeline: a first cut double-buffered one item at a
     // time and was latency-bound (~940 cycles/item measured at nnz=40);
@@ -430,6 +527,8 @@ __global__ __launch_bounds__(128) void als_solve_wave_kernel(
         }
         wave_sync();
       }
+    }
+
     }
 
     // ---- YtY base (implicit) + regularization ----
@@ -1079,6 +1178,20 @@ __global__ __launch_bounds__(128, 3) void als_woodbury2_kernel(
 // launcher
 // ---------------------------------------------------------------------------
 
+// instantiation guard: the MFMA-Gramian wave kernel exists only at F=64
+template <int FF>
+static void launch_wave_bf16g(int grid_w, hipStream_t stream,
+    const long long* indptr, const int* indices, const float* values,
+    const float* Y, const float* YtY, float* X, int n_rows, float lambda,
+    float alpha, int implicit_mode, int wr_scale, int skip, int wave_hi) {
+  if constexpr (FF == 64) {
+    hipLaunchKernelGGL((als_solve_wave_kernel<64, true>), dim3(grid_w),
+                       dim3(128), 0, stream, indptr, indices, values, Y,
+                       YtY, X, n_rows, lambda, alpha, implicit_mode,
+                       wr_scale, skip, wave_hi);
+  }
+}
+
 extern "C" void launch_als_solve(
     const long long* indptr, const int* indices, const float* values,
     const float* Y, const float* YtY, const float* V, float* X,
@@ -1109,7 +1222,15 @@ extern "C" void launch_als_solve(
   const int dense_split = e_split ? atoi(e_split) : 0;
   const int wave_hi = dense_split > 0 ? dense_split : 0x7fffffff;
 #define LAUNCH_WAVE(FF)                                                      \
-  if (dense) {                                                               \
+  if (dense && use_bf16s_g && FF == 64) {                                    \
+  launch_wave_bf16g<FF>(grid_w, stream, indptr, indices, values, Y, YtY, X, \
+                        n_rows, lambda, alpha, implicit_mode, wr_scale,      \
+                        skip, wave_hi);                                      \
+  if (dense_split > 0)                                                       \
+    hipLaunchKernelGGL((als_solve_kernel<FF>), dim3(grid_b), dim3(256), 0,   \
+                       stream, indptr, indices, values, Y, YtY, X, n_rows,   \
+                       lambda, alpha, implicit_mode, wr_scale, dense_split); \
+  } else if (dense) {                                                        \
   hipLaunchKernelGGL((als_solve_wave_kernel<FF>), dim3(grid_w), dim3(128),   \
                      0, stream, indptr, indices, values, Y, YtY, X, n_rows,  \
                      lambda, alpha, implicit_mode, wr_scale, skip, wave_hi); \
@@ -1132,6 +1253,12 @@ extern "C" void launch_als_solve(
   const char* e_bf16 = getenv("PIO_ALS_STAGE_BF16");
   const bool stage_bf16 = e_bf16 != nullptr && e_bf16[0] == '1';
   const bool use_bf16s = stage_bf16 && implicit_mode && V != nullptr;
+  // the MFMA-Gramian dense variant has its OWN opt-in: it measured
+  // SLOWER at bench shapes (the wave kernel is gather-latency-bound,
+  // so the VALU Gramian it replaces was hidden anyway) and must not
+  // ride along with the beneficial PIO_ALS_STAGE_BF16 combo
+  const char* e_mg = getenv("PIO_ALS_MFMA_GRAMIAN");
+  const bool use_bf16s_g = e_mg != nullptr && e_mg[0] == '1';
   long long wg4 = ((long long)n_rows + 3) / 4;
   int grid_w2 = (int)(wg4 < (1 << 20) ? wg4 : (1 << 20));
 #define LAUNCH_WOODBURY(FF)                                                  \

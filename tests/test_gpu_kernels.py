@@ -519,3 +519,40 @@ class TestTemplatesOnDevice:
         assert len(r.item_scores) == 5
         for s in r.item_scores:
             assert int(s.item[1:]) % 2 == 1  # category filter held
+
+
+@requires_gpu
+class TestMfmaGramian:
+    def test_dense_rows_on_matrix_cores(self):
+        """PIO_ALS_MFMA_GRAMIAN=1 routes nnz>32 rows through the MFMA
+        Gramian (bf16 inputs / fp32 accumulate) — numerics within the
+        bf16 bound vs the fp32 reference. Subprocess so the env applies
+        from the first launch."""
+        import subprocess
+        import sys
+        code = """
+import torch, math
+from predictionio_amd.ops import als as als_ops
+g = torch.Generator().manual_seed(5)
+n_rows, n_cols, f, nnz_r = 500, 800, 64, 48
+nnz = n_rows * nnz_r
+indptr = torch.arange(0, nnz + 1, nnz_r, dtype=torch.int64)[: n_rows + 1]
+ix = torch.randint(0, n_cols, (nnz,), generator=g, dtype=torch.int32)
+vv = (torch.rand(nnz, generator=g) * 3 + 0.5).float()
+Y = (torch.randn((n_cols, f), generator=g) / math.sqrt(f)).float()
+for implicit in (True, False):
+    YtY = als_ops.gramian(Y) if implicit else None
+    ref = als_ops.als_solve_ref(indptr, ix, vv, Y, YtY, lam=0.05,
+                                alpha=2.0, implicit=implicit)
+    X = als_ops.als_solve(indptr.cuda(), ix.cuda(), vv.cuda(), Y.cuda(),
+                          YtY.cuda() if implicit else None, lam=0.05,
+                          alpha=2.0, implicit=implicit).cpu()
+    rel = ((X - ref).norm() / ref.norm()).item()
+    assert rel < 5e-3, (implicit, rel)
+print("GRAMIAN_OK")
+"""
+        env = dict(os.environ, PIO_ALS_MFMA_GRAMIAN="1")
+        r = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+        assert "GRAMIAN_OK" in r.stdout
