@@ -224,19 +224,39 @@ class ALSTrainer:
 
     def load_checkpoint(self) -> int:
         """Resume local shards if a matching checkpoint exists; returns
-        the iteration to continue FROM (0 = fresh start)."""
+        the iteration to continue FROM (0 = fresh start).
+
+        Multi-rank: ranks must agree on the iteration — a crash between
+        one rank's save and another's leaves unequal checkpoints, and
+        resuming from mixed iterations would desynchronize the
+        collectives AND the math. All ranks exchange their local
+        candidate; on ANY disagreement (or any rank missing its file)
+        everyone restarts fresh."""
         path = self._ckpt_path()
-        if path is None or not os.path.exists(path):
-            return 0
-        blob = torch.load(path, weights_only=True)
-        if (blob.get("rank") != self.p.rank
-                or tuple(blob.get("u_block", ())) != (self.u_lo, self.u_hi)
-                or tuple(blob.get("i_block", ())) != (self.i_lo, self.i_hi)):
-            return 0  # shape/shard mismatch: ignore stale checkpoint
-        self.X = blob["X"].to(self.device)
-        self.Y = blob["Y"].to(self.device)
-        self._pending_y = None
-        return int(blob["iteration"])
+        it = 0
+        if path is not None and os.path.exists(path):
+            blob = torch.load(path, weights_only=True)
+            if (blob.get("rank") == self.p.rank
+                    and tuple(blob.get("u_block", ()))
+                    == (self.u_lo, self.u_hi)
+                    and tuple(blob.get("i_block", ()))
+                    == (self.i_lo, self.i_hi)):
+                it = int(blob["iteration"])
+        if pdist.is_distributed():
+            lo = -pdist.max_scalar(float(-it))   # MIN over ranks
+            hi = pdist.max_scalar(float(it))
+            if lo != hi or it == 0:
+                if it != 0:
+                    import logging
+                    logging.getLogger(__name__).warning(
+                        "ALS checkpoint iterations disagree across ranks "
+                        "(%s..%s) — restarting fresh", lo, hi)
+                return 0
+        if it > 0:
+            self.X = blob["X"].to(self.device)
+            self.Y = blob["Y"].to(self.device)
+            self._pending_y = None
+        return it
 
     def fit(self) -> Tuple[torch.Tensor, torch.Tensor]:
         """Train and return the FULL (n_users x f, n_items x f) factor

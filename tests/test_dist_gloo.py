@@ -303,3 +303,37 @@ class TestDistributedImplicit:
             t.step()
         assert torch.allclose(torch.tensor(res[0][0]), t.X,
                               atol=1e-4, rtol=1e-4)
+
+
+def _ckpt_disagreement(rank, world):
+    """Mixed checkpoint state across ranks must resolve to a FRESH
+    start on every rank (mixed iterations would desynchronize both the
+    collectives and the math)."""
+    import tempfile
+
+    from predictionio_amd.models.als import ALSParams, ALSTrainer
+    d = os.environ["PIO_TEST_CKPT_DIR"]
+    p = ALSParams(rank=16, iterations=3, seed=1, checkpoint_every=1,
+                  checkpoint_dir=d)
+    t = ALSTrainer(p, 20, 12, torch.device("cpu"))
+    g = torch.Generator().manual_seed(3)
+    users = torch.randint(0, 20, (80,), generator=g, dtype=torch.int32)
+    items = torch.randint(0, 12, (80,), generator=g, dtype=torch.int32)
+    t.set_ratings(users, items, torch.ones(80))
+    t.init_factors()
+    t.step()  # collective — all ranks must participate
+    if rank == 0:
+        # only rank 0 persists (simulates a crash mid-save)
+        t.save_checkpoint(2)
+    import torch.distributed as dist
+    dist.barrier()
+    start = t.load_checkpoint()
+    assert start == 0, f"rank {rank} resumed from {start}"
+    return True
+
+
+class TestCheckpointAgreement:
+    def test_disagreement_restarts_fresh(self, tmp_path, monkeypatch):
+        monkeypatch.setenv("PIO_TEST_CKPT_DIR", str(tmp_path))
+        res = _spawn("_ckpt_disagreement", port=29641)
+        assert res[0] is True and res[1] is True
