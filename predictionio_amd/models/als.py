@@ -169,6 +169,14 @@ class ALSTrainer:
         timing = os.environ.get("PIO_PHASE_TIMES") == "1"
         implicit = self.p.implicit
         t0 = self._tsync() if timing else 0.0
+        # YtY from the LOCAL shard + an FxF all-reduce (the post-gather
+        # Gramian would redo a world-size-times-larger GEMM on every
+        # rank); issued ASYNC before the factor-gather wait so the tiny
+        # reduce completes under the big collective
+        yty = yty_w = None
+        if implicit:
+            yty, yty_w = pdist.all_reduce_sum_async(
+                als_ops.gramian(self.Y))
         # full Y for the user half: consume the gather pipelined from the
         # previous iteration's item half (first iteration gathers fresh)
         if self._pending_y is not None:
@@ -179,17 +187,21 @@ class ALSTrainer:
                     else None)
             Yfull = pdist.all_gather_rows(self.Y, self.n_items,
                                           wire_dtype=wire)
-        # YtY from the LOCAL shard + an FxF all-reduce: the post-gather
-        # Gramian would redo a world-size-times-larger GEMM on every rank
-        yty = (pdist.all_reduce_sum(als_ops.gramian(self.Y))
-               if implicit else None)
+        if yty_w is not None:
+            yty_w.wait()
         t1 = self._tsync() if timing else 0.0
         self.X, gx = self._solve_half(
             self.user_csr, Yfull, yty, self.u_hi - self.u_lo, self.n_users)
         t2 = self._tsync() if timing else 0.0
+        yty2 = yty2_w = None
+        if implicit:
+            # self.X is complete once the solves above are stream-ordered
+            # done; overlap its Gramian reduce with the X-gather wait
+            yty2, yty2_w = pdist.all_reduce_sum_async(
+                als_ops.gramian(self.X))
         Xfull = gx.finish()
-        yty2 = (pdist.all_reduce_sum(als_ops.gramian(self.X))
-                if implicit else None)
+        if yty2_w is not None:
+            yty2_w.wait()
         t3 = self._tsync() if timing else 0.0
         self.Y, self._pending_y = self._solve_half(
             self.item_csr, Xfull, yty2, self.i_hi - self.i_lo, self.n_items)

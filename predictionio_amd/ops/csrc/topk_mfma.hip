@@ -72,8 +72,13 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
   return false;
 }
 
+// QB = query blocks per workgroup (each 64 queries, scored SEQUENTIALLY
+// per staged chunk). QB=2 amortizes the staging phase (the round-2 probe
+// measured ~50/50 stage/score) over 2x the MFMA work and HALVES the
+// total HBM item traffic (half as many ublocks read the full Y set);
+// the cost is 2x list LDS (occupancy ~7 -> ~5 WGs/CU) and 2x epilogue.
 template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK,
-          int NWAVES = TM_WAVES, bool GLL = false>
+          int NWAVES = TM_WAVES, bool GLL = false, int QB = 1>
 __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
@@ -97,9 +102,10 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
       lds_raw + (DBUF ? 2 : 1) * CH * ROWB);
   const int KP = K + 1;  // stride coprime with the 32 banks (v3 lesson)
   constexpr int BS = NWAVES * 64;        // block size
-  constexpr int UPB = NWAVES * TM_QPW;   // queries per block
-  int* topi = reinterpret_cast<int*>(topv + UPB * KP);
-  float* th_lds = reinterpret_cast<float*>(topi + UPB * KP);
+  constexpr int UPB = NWAVES * TM_QPW;   // queries per query block
+  constexpr int UPBT = UPB * QB;         // queries per workgroup
+  int* topi = reinterpret_cast<int*>(topv + UPBT * KP);
+  float* th_lds = reinterpret_cast<float*>(topi + UPBT * KP);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -111,44 +117,53 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
   unsigned long long pt = 0, acc_setup = 0, acc_stage = 0, acc_score = 0;
   if (probe) pt = wall_clock64();
 
-  const long long u0 = (long long)blockIdx.x * UPB;
-  const long long guser = u0 + wave * TM_QPW + lq;
-  const bool has_user = guser < B;
+  const long long u0 = (long long)blockIdx.x * UPBT;
+  long long guser[QB];
+  bool has_user[QB];
+#pragma unroll
+  for (int qb = 0; qb < QB; ++qb) {
+    guser[qb] = u0 + qb * UPB + wave * TM_QPW + lq;
+    has_user[qb] = guser[qb] < B;
+  }
   const int slice = blockIdx.y;
   const long long per = (N + n_slices - 1) / n_slices;
   const long long it0 = (long long)slice * per;
   const long long it1 = min(N, it0 + per);
 
   // ---- X fragments: one b128 per K-step, held for the whole kernel
-  bf16x8 xf[KS];
+  bf16x8 xf[QB][KS];
 #pragma unroll
-  for (int ks = 0; ks < KS; ++ks) {
-    if (has_user) {
-      xf[ks] = *reinterpret_cast<const bf16x8*>(
-          &Xq[guser * F + ks * 32 + lg * 8]);
-    } else {
-      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-      xf[ks] = z;
+  for (int qb = 0; qb < QB; ++qb)
+#pragma unroll
+    for (int ks = 0; ks < KS; ++ks) {
+      if (has_user[qb]) {
+        xf[qb][ks] = *reinterpret_cast<const bf16x8*>(
+            &Xq[guser[qb] * F + ks * 32 + lg * 8]);
+      } else {
+        bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        xf[qb][ks] = z;
+      }
     }
-  }
 
   // ---- init the per-query top-K lists + shared thresholds
-  for (int e = tid; e < UPB * KP; e += BS) {
+  for (int e = tid; e < UPBT * KP; e += BS) {
     topv[e] = -FLT_MAX;
     topi[e] = -1;
   }
-  if (tid < UPB) th_lds[tid] = -FLT_MAX;
+  if (tid < UPBT) th_lds[tid] = -FLT_MAX;
   __syncthreads();
-  const int mylist = wave * TM_QPW + lq;   // this lane's query list
-  float* tvu = topv + mylist * KP;
-  int* tiu = topi + mylist * KP;
 
-  const int* ban = nullptr;
-  int bn = 0;
-  if (ban_indptr != nullptr && has_user) {
-    const long long b0 = ban_indptr[guser];
-    bn = (int)(ban_indptr[guser + 1] - b0);
-    ban = ban_indices + b0;
+  const int* ban[QB];
+  int bn[QB];
+#pragma unroll
+  for (int qb = 0; qb < QB; ++qb) {
+    ban[qb] = nullptr;
+    bn[qb] = 0;
+    if (ban_indptr != nullptr && has_user[qb]) {
+      const long long b0 = ban_indptr[guser[qb]];
+      bn[qb] = (int)(ban_indptr[guser[qb] + 1] - b0);
+      ban[qb] = ban_indices + b0;
+    }
   }
   if (probe) {
     const unsigned long long now = wall_clock64();
@@ -254,7 +269,13 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
       pt = now;
     }
 
-    // ---- MFMA: every wave scores the whole chunk for its 16 queries
+    // ---- MFMA: every wave scores the whole chunk for its queries,
+    // one 64-query block at a time (QB passes over the staged tile)
+#pragma unroll
+    for (int qb = 0; qb < QB; ++qb) {
+    const int mylist = qb * UPB + wave * TM_QPW + lq;
+    float* tvu = topv + mylist * KP;
+    int* tiu = topi + mylist * KP;
     f32x4 acc[IFR];
 #pragma unroll
     for (int i = 0; i < IFR; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -268,7 +289,7 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
             reinterpret_cast<const char*>(yb) +
             row * ROWB + (col ^ ((row & SWM) << 4)));
         acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, xf[ks], acc[i], 0, 0, 0);
+            a, xf[qb][ks], acc[i], 0, 0, 0);
       }
     }
 
@@ -279,7 +300,7 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) mymax = fmaxf(mymax, acc[i][r]);
     const float th_b = th_lds[mylist];
-    if (has_user && mymax > th_b) {
+    if (has_user[qb] && mymax > th_b) {
       // serialize the query's 4 lane groups: exec-masked blocks of one
       // wave run in program order, and within a group the active lanes
       // all own different queries, so list writes never contend.
@@ -306,8 +327,9 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
               const float s = acc[i][r];
               if (s > th && li < lim) {
                 if ((item_mask == nullptr || !item_mask[item]) &&
-                    (ban == nullptr ||
-                     !tm_in_sorted(ban, bn, (int)(item + item_base)))) {
+                    (ban[qb] == nullptr ||
+                     !tm_in_sorted(ban[qb], bn[qb],
+                                   (int)(item + item_base)))) {
                   int mi = 0;
                   float mv = tvu[0];
                   for (int q = 1; q < K; ++q)
@@ -325,6 +347,7 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
         }
       }
     }
+    }  // qb
     if (probe) {
       const unsigned long long now = wall_clock64();
       acc_score += now - pt;
@@ -349,7 +372,7 @@ __global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
   if (probe) pt = wall_clock64();
 
   // ---- write out: one candidate group per slice per query
-  for (int e = tid; e < UPB * K; e += BS) {
+  for (int e = tid; e < UPBT * K; e += BS) {
     const int list = e / K;
     const int q = e % K;
     const long long gu = u0 + list;
@@ -406,11 +429,16 @@ extern "C" void launch_topk_mfma(
   const char* e_w = getenv("PIO_TOPK_WIDE");
   const bool use_wide = e_w != nullptr && e_w[0] == '1' && f >= 64 &&
                         prof == nullptr && !use_db && chunk == TM_CHUNK;
+  const char* e_q = getenv("PIO_TOPK_QB");
+  const bool use_qb2 = e_q != nullptr && e_q[0] == '2' && !use_db &&
+                       !use_gll && !use_wide && chunk == TM_CHUNK;
+  dim3 grid_q2((B + 2 * TM_UPB - 1) / (2 * TM_UPB), n_slices);
 #define LAUNCH_M(FF)                                                         \
   do {                                                                       \
     size_t lds_bytes = (size_t)(use_db ? 2 : 1) * chunk * (FF * 2) +         \
-                       (sizeof(float) + sizeof(int)) * TM_UPB * (K + 1) +    \
-                       sizeof(float) * TM_UPB;                               \
+                       (sizeof(float) + sizeof(int)) *                       \
+                           (use_qb2 ? 2 : 1) * TM_UPB * (K + 1) +            \
+                       sizeof(float) * (use_qb2 ? 2 : 1) * TM_UPB;           \
     static bool attr_set_##FF = false;                                       \
     if (!attr_set_##FF && lds_bytes > 64 * 1024) {                           \
       hipFuncSetAttribute(                                                   \
@@ -419,9 +447,34 @@ extern "C" void launch_topk_mfma(
       hipFuncSetAttribute(                                                   \
           reinterpret_cast<const void*>(&topk_mfma_kernel<FF, true, false>),        \
           hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(                                     \
+              &topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES,        \
+                                false, 2>),                                  \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(                                     \
+              &topk_mfma_kernel<FF, true, false, TM_CHUNK, TM_WAVES,         \
+                                false, 2>),                                  \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);           \
       attr_set_##FF = true;                                                  \
     }                                                                        \
-    if (prof != nullptr) {                                                   \
+    if (use_qb2) {                                                           \
+      if (prof != nullptr)                                                   \
+        hipLaunchKernelGGL(                                                  \
+            (topk_mfma_kernel<FF, true, false, TM_CHUNK, TM_WAVES, false,    \
+                              2>),                                           \
+            grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
+            ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
+            prof);                                                           \
+      else                                                                   \
+        hipLaunchKernelGGL(                                                  \
+            (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, false,   \
+                              2>),                                           \
+            grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
+            ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
+            nullptr);                                                        \
+    } else if (prof != nullptr) {                                            \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, true>), grid, block,  \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \

@@ -156,7 +156,8 @@ def _topk_score_mfma(Xq: torch.Tensor, Y: torch.Tensor, K: int,
         # insert volume grows ~linearly with n_slices (each slice resets
         # the per-query threshold), so use just enough slices to fill
         # the chip (measured sweep: scripts/mfma_phase_probe.py)
-        ublocks = (B + 63) // 64
+        upw = 128 if os.environ.get("PIO_TOPK_QB") == "2" else 64
+        ublocks = (B + upw - 1) // upw
         n_slices = max(2, min(1536 // ublocks, (N + 255) // 256))
     # Y's bf16 copy is cached (factors are static across serving batches);
     # Xq is cast per call — it changes every batch, and under hipGraph

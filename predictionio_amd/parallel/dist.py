@@ -234,6 +234,16 @@ def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+def all_reduce_sum_async(t: torch.Tensor):
+    """Issue an in-place SUM all-reduce without waiting; returns
+    (tensor, work_or_None). Lets the tiny FxF Gramian reduce complete
+    under the big factor-gather wait in the ALS step (NOTES.md future
+    work item 3); caller must work.wait() before reading the tensor."""
+    if not is_distributed():
+        return t, None
+    return t, dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=True)
+
+
 def barrier() -> None:
     if is_distributed():
         dist.barrier()
