@@ -77,9 +77,16 @@ __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
 // measured ~50/50 stage/score) over 2x the MFMA work and HALVES the
 // total HBM item traffic (half as many ublocks read the full Y set);
 // the cost is 2x list LDS (occupancy ~7 -> ~5 WGs/CU) and 2x epilogue.
+// QB=2 needs ~2x the live registers (both query blocks' X fragments +
+// insert state): launch_bounds min-waves 6 would cap it at 80 VGPRs and
+// spill the hot loop to scratch (measured 3-4x slower — scratch counts
+// toward vmcnt and drains the staging pipeline every chunk), so QB=2
+// instantiations get min-waves 4 (128 VGPRs; LDS caps occupancy at
+// ~5 WGs/CU for K=20 anyway).
 template <int F, bool PROF, bool DBUF, int CH = TM_CHUNK,
           int NWAVES = TM_WAVES, bool GLL = false, int QB = 1>
-__global__ __launch_bounds__(NWAVES * 64, 24 / NWAVES) void topk_mfma_kernel(
+__global__ __launch_bounds__(NWAVES * 64, QB == 1 ? 24 / NWAVES : 4)
+void topk_mfma_kernel(
     const unsigned short* __restrict__ Xq,   // B x F bf16
     const unsigned short* __restrict__ Y,    // N x F bf16
     const uint8_t* __restrict__ item_mask,   // N or nullptr
