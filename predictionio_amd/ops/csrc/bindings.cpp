@@ -29,7 +29,7 @@ extern "C" void launch_topk_mfma(
     const uint8_t* item_mask, const long long* ban_indptr,
     const int* ban_indices, float* out_val, int* out_idx,
     int B, long long N, int f, int K, int n_slices, int item_base,
-    unsigned long long* prof, hipStream_t stream);
+    unsigned long long* prof, unsigned* th_g, hipStream_t stream);
 
 namespace {
 
@@ -243,12 +243,26 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score_mfma(
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream =
       c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  // cross-slice threshold exchange: a [B] tm_enc-coded atomicMax cell
+  // per query, zero-init (= below every real score), shared by all of a
+  // query's slice WGs through L2. Auto-on for B >= 512 (+14% at B=4096);
+  // at small B the whole grid hammers a few cells and the atomic queue
+  // serializes (B=1 measured 4x slower — profiles/serve_gth_ab_r2.log).
+  // PIO_TOPK_GTH=0 forces off, =1 forces on.
+  torch::Tensor th_g;
+  unsigned* th_g_ptr = nullptr;
+  const char* e_gth = getenv("PIO_TOPK_GTH");
+  const bool gth = e_gth != nullptr ? (e_gth[0] == '1') : (B >= 512);
+  if (gth) {
+    th_g = torch::zeros({B}, Xq.options().dtype(torch::kInt32));
+    th_g_ptr = reinterpret_cast<unsigned*>(th_g.data_ptr<int>());
+  }
   launch_topk_mfma(
       reinterpret_cast<const unsigned short*>(Xq.data_ptr()),
       reinterpret_cast<const unsigned short*>(Y.data_ptr()), mask_ptr,
       bi_ptr, bx_ptr, out_val.data_ptr<float>(), out_idx.data_ptr<int>(),
       (int)B, (long long)N, (int)f, (int)K, (int)n_slices, (int)item_base,
-      prof_ptr, stream);
+      prof_ptr, th_g_ptr, stream);
   C10_HIP_CHECK(hipGetLastError());
   return {out_val, out_idx};
 }

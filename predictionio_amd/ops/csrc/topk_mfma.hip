@@ -61,6 +61,17 @@ typedef __attribute__((ext_vector_type(8))) unsigned short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
+// Order-preserving float<->u32 map for atomicMax on scores (classic
+// sign-flip transform: negative floats map to ~bits, non-negative to
+// bits|0x80000000, making unsigned order == float order).
+__device__ __forceinline__ unsigned tm_enc(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+__device__ __forceinline__ float tm_dec(unsigned k) {
+  return __uint_as_float((k & 0x80000000u) ? (k & 0x7FFFFFFFu) : ~k);
+}
+
 __device__ __forceinline__ bool tm_in_sorted(const int* arr, int n, int x) {
   int lo = 0, hi = n - 1;
   while (lo <= hi) {
@@ -95,7 +106,12 @@ void topk_mfma_kernel(
     float* __restrict__ out_val,             // B x n_slices x K
     int* __restrict__ out_idx,
     int B, long long N, int K, int n_slices, int item_base,
-    unsigned long long* prof)
+    unsigned long long* prof,
+    // optional [B] cross-slice threshold (tm_enc-coded, init 0): the
+    // concurrently-running WGs of a query's OTHER slices publish their
+    // K-th-best through L2, so each slice stops re-paying the full
+    // insert ramp from -inf (insert volume was ~linear in n_slices)
+    unsigned* __restrict__ th_g)
 {
   constexpr int ROWB = F * 2;            // bytes per staged Y row
   constexpr int SWM = (F >= 64) ? 7 : 3; // XOR-swizzle row mask
@@ -264,6 +280,20 @@ void topk_mfma_kernel(
     } else if (!DBUF) {
       __syncthreads();  // all waves done reading the previous chunk
       drain_to(yb);
+      // fold the cross-slice global threshold into the local one every
+      // 8 chunks (one L2-hot load per query; each wave's lg==0 lanes
+      // update only their own wave's lists, and the barrier below
+      // orders the write before the epilogue's reads)
+      if (th_g != nullptr && (ci & 7) == 0 && lg == 0) {
+#pragma unroll
+        for (int qb = 0; qb < QB; ++qb) {
+          if (has_user[qb]) {
+            const int ml = qb * UPB + wave * TM_QPW + lq;
+            const float gv = tm_dec(th_g[guser[qb]]);
+            if (gv > th_lds[ml]) th_lds[ml] = gv;
+          }
+        }
+      }
       __syncthreads();
       // issue the NEXT chunk's global loads AFTER the barrier (a
       // __syncthreads compiles to s_waitcnt vmcnt(0)) so they fly
@@ -347,6 +377,9 @@ void topk_mfma_kernel(
                   for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
                   th = nm;
                   th_lds[mylist] = nm;
+                  // publish to the other slices once the list is full
+                  if (th_g != nullptr && nm > -FLT_MAX)
+                    atomicMax(&th_g[guser[qb]], tm_enc(nm));
                 }
               }
             }
@@ -407,13 +440,14 @@ static void launch_topk_mfma_wide(
     const unsigned short* Xq, const unsigned short* Y,
     const uint8_t* item_mask, const long long* ban_indptr,
     const int* ban_indices, float* out_val, int* out_idx,
-    int B, long long N, int K, int n_slices, int item_base) {
+    int B, long long N, int K, int n_slices, int item_base,
+    unsigned* th_g) {
   if constexpr (FF >= 64) {
     hipLaunchKernelGGL(
         (topk_mfma_kernel<FF, false, false, TM_CHUNK, 8>), grid,
         dim3(512), lds_bytes, stream, Xq, Y, item_mask, ban_indptr,
         ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,
-        nullptr);
+        nullptr, th_g);
   }
 }
 
@@ -422,7 +456,7 @@ extern "C" void launch_topk_mfma(
     const uint8_t* item_mask, const long long* ban_indptr,
     const int* ban_indices, float* out_val, int* out_idx,
     int B, long long N, int f, int K, int n_slices, int item_base,
-    unsigned long long* prof, hipStream_t stream)
+    unsigned long long* prof, unsigned* th_g, hipStream_t stream)
 {
   dim3 grid((B + TM_UPB - 1) / TM_UPB, n_slices);
   dim3 block(256);
@@ -473,59 +507,59 @@ extern "C" void launch_topk_mfma(
                               2>),                                           \
             grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            prof);                                                           \
+            prof, th_g);                                                           \
       else                                                                   \
         hipLaunchKernelGGL(                                                  \
             (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, false,   \
                               2>),                                           \
             grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            nullptr);                                                        \
+            nullptr, th_g);                                                        \
     } else if (prof != nullptr) {                                            \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, true>), grid, block,  \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, prof);                                 \
+                           item_base, prof, th_g);                                 \
       else if (chunk == 128)                                                 \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false, 128>), grid,   \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, prof);                    \
+                           K, n_slices, item_base, prof, th_g);                    \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false>), grid, block, \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, prof);                                 \
+                           item_base, prof, th_g);                                 \
     } else if (use_wide) {                                                   \
       size_t lds_w = (size_t)TM_CHUNK * (FF * 2) +                           \
                      (sizeof(float) + sizeof(int)) * 128 * (K + 1) +         \
                      sizeof(float) * 128;                                    \
       launch_topk_mfma_wide<FF>(grid_w, lds_w, stream, Xq, Y, item_mask,     \
                                 ban_indptr, ban_indices, out_val, out_idx,   \
-                                B, N, K, n_slices, item_base);               \
+                                B, N, K, n_slices, item_base, th_g);         \
     } else {                                                                 \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, true>), grid, block, \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, nullptr);                              \
+                           item_base, nullptr, th_g);                              \
       else if (chunk == 128)                                                 \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false, 128>), grid,  \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, nullptr);                 \
+                           K, n_slices, item_base, nullptr, th_g);                 \
       else if (use_gll)                                                      \
         hipLaunchKernelGGL(                                                  \
             (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, true>),  \
             grid, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            nullptr);                                                        \
+            nullptr, th_g);                                                        \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false>), grid,       \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, nullptr);                 \
+                           K, n_slices, item_base, nullptr, th_g);                 \
     }                                                                        \
   } while (0)
   switch (f) {

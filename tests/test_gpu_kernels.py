@@ -192,6 +192,29 @@ class TestTopKKernel:
         # bf16-near-tie at the candidate cut
         assert torch.allclose(gv, rv, atol=5e-3, rtol=5e-3)
 
+    @pytest.mark.parametrize("gth", ["0", "1"])
+    def test_mfma_gth_modes_match_fp32(self, gth, monkeypatch):
+        """Cross-slice global-threshold exchange (PIO_TOPK_GTH) on AND
+        off must both reproduce the fp32 reference: the global cell only
+        prunes items already beaten by K better ones somewhere, so the
+        merged top-K is unchanged (many slices so pruning actually
+        fires)."""
+        from predictionio_amd.ops import topk as topk_ops
+        monkeypatch.setenv("PIO_TOPK_GTH", gth)
+        g = torch.Generator().manual_seed(11)
+        B, N, K, f = 96, 300_000, 20, 64
+        Xq = torch.randn((B, f), generator=g).float()
+        Y = torch.randn((N, f), generator=g).float()
+        rv, ri = topk_ops.topk_score_ref(Xq, Y, K)
+        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K, mode="mfma",
+                                     n_slices=96)
+        gv, gi = gv.cpu(), gi.cpu()
+        hits = sum(len(set(gi[b].tolist()) & set(ri[b].tolist()))
+                   for b in range(B))
+        assert hits / (B * K) >= 0.999
+        chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
+        assert torch.allclose(gv, chosen, atol=1e-5, rtol=1e-5)
+
     def test_mfma_more_k_than_items(self):
         from predictionio_amd.ops import topk as topk_ops
         Xq = torch.randn((3, 64)).float().cuda()
