@@ -243,16 +243,17 @@ std::tuple<torch::Tensor, torch::Tensor> topk_score_mfma(
   c10::hip::HIPGuardMasqueradingAsCUDA guard(Y.device());
   hipStream_t stream =
       c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  // cross-slice threshold exchange: a [B] tm_enc-coded atomicMax cell
-  // per query, zero-init (= below every real score), shared by all of a
-  // query's slice WGs through L2. Auto-on for B >= 512 (+14% at B=4096);
-  // at small B the whole grid hammers a few cells and the atomic queue
-  // serializes (B=1 measured 4x slower — profiles/serve_gth_ab_r2.log).
-  // PIO_TOPK_GTH=0 forces off, =1 forces on.
+  // cross-slice threshold exchange (default ON; PIO_TOPK_GTH=0
+  // disables): a [B] tm_enc-coded atomicMax cell per query, zero-init
+  // (= below every real score), shared by all of a query's slice WGs
+  // through L2. +17-29% for B in [64, 4096]; the kernel's
+  // test-and-test-and-set read keeps the atomic queue short (the
+  // unfiltered version serialized at small B —
+  // profiles/serve_gth_ab_r2.log).
   torch::Tensor th_g;
   unsigned* th_g_ptr = nullptr;
   const char* e_gth = getenv("PIO_TOPK_GTH");
-  const bool gth = e_gth != nullptr ? (e_gth[0] == '1') : (B >= 512);
+  const bool gth = !(e_gth != nullptr && e_gth[0] == '0');
   if (gth) {
     th_g = torch::zeros({B}, Xq.options().dtype(torch::kInt32));
     th_g_ptr = reinterpret_cast<unsigned*>(th_g.data_ptr<int>());

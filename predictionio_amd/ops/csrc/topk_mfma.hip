@@ -377,8 +377,12 @@ void topk_mfma_kernel(
                   for (int q = 1; q < K; ++q) nm = fminf(nm, tvu[q]);
                   th = nm;
                   th_lds[mylist] = nm;
-                  // publish to the other slices once the list is full
-                  if (th_g != nullptr && nm > -FLT_MAX)
+                  // publish to the other slices once the list is full;
+                  // test-and-test-and-set: the plain read (L2-hot)
+                  // filters publishes already beaten by another slice,
+                  // keeping the atomic queue short
+                  if (th_g != nullptr && nm > -FLT_MAX &&
+                      tm_enc(nm) > th_g[guser[qb]])
                     atomicMax(&th_g[guser[qb]], tm_enc(nm));
                 }
               }
