@@ -261,6 +261,14 @@ void topk_mfma_kernel(
           reinterpret_cast<unsigned int*>(dst), 16, 0, 0);
     }
   };
+  // pipelined cross-slice threshold: the register holds the value read
+  // one fold period ago (tm_enc-coded; 0 decodes below every real
+  // score, so the first fold is a no-op)
+  unsigned gth_pipe[QB];
+#pragma unroll
+  for (int qb = 0; qb < QB; ++qb)
+    gth_pipe[qb] = (th_g != nullptr && lg == 0 && has_user[qb])
+        ? th_g[guser[qb]] : 0u;
   const int n_chunks = (int)((it1 - it0 + CH - 1) / CH);
   if (!GLL) load_stg(it0);
   if (DBUF) {
@@ -281,16 +289,20 @@ void topk_mfma_kernel(
       __syncthreads();  // all waves done reading the previous chunk
       drain_to(yb);
       // fold the cross-slice global threshold into the local one every
-      // 8 chunks (one L2-hot load per query; each wave's lg==0 lanes
-      // update only their own wave's lists, and the barrier below
-      // orders the write before the epilogue's reads)
+      // 8 chunks; each wave's lg==0 lanes update only their own wave's
+      // lists, and the barrier below orders the write before the
+      // epilogue's reads. PIPELINED: consume the value LOADED 8 CHUNKS
+      // AGO (a barrier since then already paid its vmcnt, so the fold
+      // never stalls on the load) and issue the next one; thresholds
+      // are monotonic so an 8-chunk-stale value only prunes less.
       if (th_g != nullptr && (ci & 7) == 0 && lg == 0) {
 #pragma unroll
         for (int qb = 0; qb < QB; ++qb) {
           if (has_user[qb]) {
             const int ml = qb * UPB + wave * TM_QPW + lq;
-            const float gv = tm_dec(th_g[guser[qb]]);
+            const float gv = tm_dec(gth_pipe[qb]);
             if (gv > th_lds[ml]) th_lds[ml] = gv;
+            gth_pipe[qb] = th_g[guser[qb]];
           }
         }
       }
