@@ -111,7 +111,8 @@ void topk_mfma_kernel(
     // concurrently-running WGs of a query's OTHER slices publish their
     // K-th-best through L2, so each slice stops re-paying the full
     // insert ramp from -inf (insert volume was ~linear in n_slices)
-    unsigned* __restrict__ th_g)
+    unsigned* __restrict__ th_g,
+    int kflags)  // bit 0: disable the ban bloom (A/B hook)
 {
   constexpr int ROWB = F * 2;            // bytes per staged Y row
   constexpr int SWM = (F >= 64) ? 7 : 3; // XOR-swizzle row mask
@@ -178,14 +179,29 @@ void topk_mfma_kernel(
 
   const int* ban[QB];
   int bn[QB];
+  // per-query 64-bit bloom over the ban list: the insert path's binary
+  // search is ~5 DEPENDENT global loads inside the serialized
+  // lane-group loop, so a one-register membership pre-test that skips
+  // it for definitely-unbanned items pays for itself after a handful
+  // of inserts (~47% false-positive at 30 bans; saturates harmlessly
+  // for huge lists). Built once at setup from this lane's own list.
+  unsigned long long bloom[QB];
 #pragma unroll
   for (int qb = 0; qb < QB; ++qb) {
     ban[qb] = nullptr;
     bn[qb] = 0;
+    bloom[qb] = 0ull;
     if (ban_indptr != nullptr && has_user[qb]) {
       const long long b0 = ban_indptr[guser[qb]];
       bn[qb] = (int)(ban_indptr[guser[qb] + 1] - b0);
       ban[qb] = ban_indices + b0;
+      if (kflags & 1) {
+        bloom[qb] = ~0ull;  // degenerate: every check hits, old path
+      } else {
+        for (int t = 0; t < bn[qb]; ++t)
+          bloom[qb] |= 1ull
+              << ((unsigned)((unsigned)ban[qb][t] * 2654435761u) >> 26);
+      }
     }
   }
   if (probe) {
@@ -375,10 +391,14 @@ void topk_mfma_kernel(
               const long long item = base + li;
               const float s = acc[i][r];
               if (s > th && li < lim) {
+                const int gitem = (int)(item + item_base);
+                const bool may_ban =
+                    ban[qb] != nullptr &&
+                    (bloom[qb] >>
+                     ((unsigned)((unsigned)gitem * 2654435761u) >> 26)) & 1;
                 if ((item_mask == nullptr || !item_mask[item]) &&
-                    (ban[qb] == nullptr ||
-                     !tm_in_sorted(ban[qb], bn[qb],
-                                   (int)(item + item_base)))) {
+                    (!may_ban ||
+                     !tm_in_sorted(ban[qb], bn[qb], gitem))) {
                   int mi = 0;
                   float mv = tvu[0];
                   for (int q = 1; q < K; ++q)
@@ -457,13 +477,13 @@ static void launch_topk_mfma_wide(
     const uint8_t* item_mask, const long long* ban_indptr,
     const int* ban_indices, float* out_val, int* out_idx,
     int B, long long N, int K, int n_slices, int item_base,
-    unsigned* th_g) {
+    unsigned* th_g, int kflags) {
   if constexpr (FF >= 64) {
     hipLaunchKernelGGL(
         (topk_mfma_kernel<FF, false, false, TM_CHUNK, 8>), grid,
         dim3(512), lds_bytes, stream, Xq, Y, item_mask, ban_indptr,
         ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,
-        nullptr, th_g);
+        nullptr, th_g, kflags);
   }
 }
 
@@ -486,6 +506,8 @@ extern "C" void launch_topk_mfma(
   const char* e_w = getenv("PIO_TOPK_WIDE");
   const bool use_wide = e_w != nullptr && e_w[0] == '1' && f >= 64 &&
                         prof == nullptr && !use_db && chunk == TM_CHUNK;
+  const char* e_bl = getenv("PIO_TOPK_BLOOM");
+  const int kflags = (e_bl != nullptr && e_bl[0] == '0') ? 1 : 0;
   const char* e_q = getenv("PIO_TOPK_QB");
   const bool use_qb2 = e_q != nullptr && e_q[0] == '2' && !use_db &&
                        !use_gll && !use_wide && chunk == TM_CHUNK;
@@ -523,59 +545,60 @@ extern "C" void launch_topk_mfma(
                               2>),                                           \
             grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            prof, th_g);                                                           \
+            prof, th_g, kflags);                                                           \
       else                                                                   \
         hipLaunchKernelGGL(                                                  \
             (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, false,   \
                               2>),                                           \
             grid_q2, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr, \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            nullptr, th_g);                                                        \
+            nullptr, th_g, kflags);                                                        \
     } else if (prof != nullptr) {                                            \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, true>), grid, block,  \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, prof, th_g);                                 \
+                           item_base, prof, th_g, kflags);                                 \
       else if (chunk == 128)                                                 \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false, 128>), grid,   \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, prof, th_g);                    \
+                           K, n_slices, item_base, prof, th_g, kflags);                    \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, true, false>), grid, block, \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, prof, th_g);                                 \
+                           item_base, prof, th_g, kflags);                                 \
     } else if (use_wide) {                                                   \
       size_t lds_w = (size_t)TM_CHUNK * (FF * 2) +                           \
                      (sizeof(float) + sizeof(int)) * 128 * (K + 1) +         \
                      sizeof(float) * 128;                                    \
       launch_topk_mfma_wide<FF>(grid_w, lds_w, stream, Xq, Y, item_mask,     \
                                 ban_indptr, ban_indices, out_val, out_idx,   \
-                                B, N, K, n_slices, item_base, th_g);         \
+                                B, N, K, n_slices, item_base, th_g,          \
+                                kflags);                                     \
     } else {                                                                 \
       if (use_db)                                                            \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, true>), grid, block, \
                            lds_bytes, stream, Xq, Y, item_mask, ban_indptr,  \
                            ban_indices, out_val, out_idx, B, N, K, n_slices, \
-                           item_base, nullptr, th_g);                              \
+                           item_base, nullptr, th_g, kflags);                              \
       else if (chunk == 128)                                                 \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false, 128>), grid,  \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, nullptr, th_g);                 \
+                           K, n_slices, item_base, nullptr, th_g, kflags);                 \
       else if (use_gll)                                                      \
         hipLaunchKernelGGL(                                                  \
             (topk_mfma_kernel<FF, false, false, TM_CHUNK, TM_WAVES, true>),  \
             grid, block, lds_bytes, stream, Xq, Y, item_mask, ban_indptr,    \
             ban_indices, out_val, out_idx, B, N, K, n_slices, item_base,     \
-            nullptr, th_g);                                                        \
+            nullptr, th_g, kflags);                                                        \
       else                                                                   \
         hipLaunchKernelGGL((topk_mfma_kernel<FF, false, false>), grid,       \
                            block, lds_bytes, stream, Xq, Y, item_mask,       \
                            ban_indptr, ban_indices, out_val, out_idx, B, N,  \
-                           K, n_slices, item_base, nullptr, th_g);                 \
+                           K, n_slices, item_base, nullptr, th_g, kflags);                 \
     }                                                                        \
   } while (0)
   switch (f) {
