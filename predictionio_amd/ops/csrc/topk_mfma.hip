@@ -195,8 +195,11 @@ void topk_mfma_kernel(
       const long long b0 = ban_indptr[guser[qb]];
       bn[qb] = (int)(ban_indptr[guser[qb] + 1] - b0);
       ban[qb] = ban_indices + b0;
-      if (kflags & 1) {
-        bloom[qb] = ~0ull;  // degenerate: every check hits, old path
+      if ((kflags & 1) || bn[qb] > 512) {
+        // disabled, or the list is big enough that the 64-bit bloom
+        // would saturate anyway — skip the build, every check falls
+        // through to the binary search (measured 0.998x at 300 bans)
+        bloom[qb] = ~0ull;
       } else {
         for (int t = 0; t < bn[qb]; ++t)
           bloom[qb] |= 1ull
