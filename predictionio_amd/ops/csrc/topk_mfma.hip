@@ -439,6 +439,19 @@ void topk_mfma_kernel(
         drain_to(ybuf(ci + 1));
         if (ci + 2 < n_chunks) load_stg(base + 2 * CH);
       }
+      // same pipelined cross-slice threshold fold as the main path
+      // (the barrier below orders the th_lds writes)
+      if (th_g != nullptr && (ci & 7) == 0 && lg == 0) {
+#pragma unroll
+        for (int qb = 0; qb < QB; ++qb) {
+          if (has_user[qb]) {
+            const int ml = qb * UPB + wave * TM_QPW + lq;
+            const float gv = tm_dec(gth_pipe[qb]);
+            if (gv > th_lds[ml]) th_lds[ml] = gv;
+            gth_pipe[qb] = th_g[guser[qb]];
+          }
+        }
+      }
       __syncthreads();
       if (probe) {
         const unsigned long long now = wall_clock64();
