@@ -215,6 +215,36 @@ class TestTopKKernel:
         chosen = (Xq @ Y.t()).gather(1, gi.clamp_min(0))
         assert torch.allclose(gv, chosen, atol=1e-5, rtol=1e-5)
 
+    @pytest.mark.parametrize("bloom", ["0", "1"])
+    def test_mfma_ban_bloom_modes(self, bloom, monkeypatch):
+        """Ban handling with the bloom pre-test on AND off: no banned
+        item may ever be served, and recall vs the fp32 reference stays
+        perfect (the bloom only SKIPS searches for items provably not
+        in the list)."""
+        from predictionio_amd.ops import topk as topk_ops
+        monkeypatch.setenv("PIO_TOPK_BLOOM", bloom)
+        g = torch.Generator().manual_seed(23)
+        B, N, K, f, nb = 128, 200_000, 20, 64, 30
+        Xq = torch.randn((B, f), generator=g).float()
+        Y = torch.randn((N, f), generator=g).float()
+        bi = torch.arange(B + 1, dtype=torch.int64) * nb
+        bxl = torch.randint(0, N, (B * nb,), generator=g,
+                            dtype=torch.int32)
+        bx = torch.sort(bxl.view(B, nb), dim=1)[0].reshape(-1)
+        rv, ri = topk_ops.topk_score_ref(Xq, Y, K, ban_indptr=bi,
+                                         ban_indices=bx)
+        gv, gi = topk_ops.topk_score(Xq.cuda(), Y.cuda(), K, mode="mfma",
+                                     ban_indptr=bi.cuda(),
+                                     ban_indices=bx.cuda())
+        gi = gi.cpu()
+        for b in range(B):
+            assert not (set(gi[b].tolist())
+                        & set(bx.view(B, nb)[b].tolist())), \
+                f"banned item served at row {b} (bloom={bloom})"
+        hits = sum(len(set(gi[b].tolist()) & set(ri[b].tolist()))
+                   for b in range(B))
+        assert hits / (B * K) >= 0.999
+
     def test_mfma_more_k_than_items(self):
         from predictionio_amd.ops import topk as topk_ops
         Xq = torch.randn((3, 64)).float().cuda()
