@@ -22,6 +22,7 @@ top-K kernel launch.
 
 from __future__ import annotations
 
+import warnings
 from collections import defaultdict
 from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Set, Tuple
@@ -272,7 +273,10 @@ class CooccurrenceAlgorithm(Algorithm):
             torch.stack([u, i]).to(device),
             torch.ones(u.numel(), device=device),
             (n_users, n_items)).coalesce()
-        C = torch.sparse.mm(A.t(), A).coalesce()
+        with warnings.catch_warnings():
+            # torch's own "sparse CSR is beta" notice, not actionable
+            warnings.simplefilter("ignore", UserWarning)
+            C = torch.sparse.mm(A.t(), A).coalesce()
         ii = C.indices()[0].cpu().numpy()
         jj = C.indices()[1].cpu().numpy()
         vv = C.values().cpu().numpy()
