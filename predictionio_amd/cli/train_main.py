@@ -28,8 +28,13 @@ def main():
     import torch
     from predictionio_amd.parallel import dist as pdist
     rank, world = pdist.init_from_env()
-    if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    lr = int(os.environ.get("LOCAL_RANK", "0"))
+    if torch.cuda.is_available() and lr < torch.cuda.device_count():
+        # only bind a device when this rank actually has one (with more
+        # ranks than GPUs init_from_env already fell back to CPU/gloo)
+        if torch.cuda.device_count() >= int(
+                os.environ.get("LOCAL_WORLD_SIZE", str(world))):
+            torch.cuda.set_device(lr)
 
     path = args.variant if os.path.isabs(args.variant) \
         else os.path.join(d, args.variant)

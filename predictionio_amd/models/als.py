@@ -59,9 +59,7 @@ class ALSTrainer:
         self.p = params
         self.n_users = n_users
         self.n_items = n_items
-        self.device = device or (
-            torch.device("cuda") if torch.cuda.is_available()
-            else torch.device("cpu"))
+        self.device = device or pdist.compute_device()
         world, rank = pdist.get_world_size(), pdist.get_rank()
         self.u_lo, self.u_hi = pdist.block_bounds(n_users, world, rank)
         self.i_lo, self.i_hi = pdist.block_bounds(n_items, world, rank)

@@ -41,12 +41,40 @@ def init_from_env(backend: str | None = None,
     if world <= 1:
         return 0, 1
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # nccl(=RCCL) needs one DISTINCT device per local rank; with
+        # more ranks than GPUs (e.g. `pio train --gpus 2` on a 1-GPU
+        # box) fall back to CPU/gloo instead of crashing rank >= count
+        local_world = int(os.environ.get("LOCAL_WORLD_SIZE", str(world)))
+        n_gpus = torch.cuda.device_count() if torch.cuda.is_available() \
+            else 0
+        if n_gpus >= local_world:
+            backend = "nccl"
+        else:
+            backend = "gloo"
+            if n_gpus > 0:
+                import logging
+                logging.getLogger(__name__).warning(
+                    "%d ranks but only %d visible GPU(s) — training on "
+                    "CPU over gloo; use --gpus <= GPU count for RCCL",
+                    local_world, n_gpus)
     dist.init_process_group(
         backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
     if backend == "nccl":
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     return get_rank(), get_world_size()
+
+
+def compute_device() -> torch.device:
+    """Device the training compute should live on: cuda when a GPU is
+    visible AND the collective backend can carry device tensors (nccl,
+    or not distributed at all). When init_from_env fell back to gloo
+    (more ranks than GPUs), compute follows the collectives to the CPU —
+    gloo cannot all-gather CUDA tensors."""
+    if not torch.cuda.is_available():
+        return torch.device("cpu")
+    if is_distributed() and dist.get_backend() != "nccl":
+        return torch.device("cpu")
+    return torch.device("cuda")
 
 
 def block_bounds(n: int, world: int, rank: int) -> Tuple[int, int]:

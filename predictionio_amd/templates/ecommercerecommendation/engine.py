@@ -30,6 +30,8 @@ from typing import Any, Dict, List, Optional, Set
 
 import torch
 
+from predictionio_amd.parallel import dist as pdist
+
 from predictionio_amd.controller import (
     Algorithm, DataSource as BaseDataSource, Engine, EngineFactory,
     Preparator as BasePreparator, SanityCheck, Serving as BaseServing,
@@ -263,8 +265,7 @@ class ECommAlgorithm(Algorithm):
             lambda_=float(self.params.get("lambda", 0.01)),
             alpha=float(self.params.get("alpha", 1.0)),
             implicit=True, seed=self.params.get("seed"))
-        device = torch.device("cuda") if torch.cuda.is_available() \
-            else torch.device("cpu")
+        device = pdist.compute_device()
         X, Y = train_als(users, items, vals, len(user_map), len(item_map),
                          p, device=device)
         # popularity = buy counts (trainDefault :207-246)

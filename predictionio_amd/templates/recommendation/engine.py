@@ -28,6 +28,8 @@ from typing import Any, Dict, List, Optional, Tuple
 
 import torch
 
+from predictionio_amd.parallel import dist as pdist
+
 from predictionio_amd.controller import (
     Algorithm, DataSource as BaseDataSource, Engine, EngineFactory, Params,
     PersistentModel, Preparator as BasePreparator, SanityCheck, Serving as
@@ -279,8 +281,7 @@ class ALSAlgorithm(Algorithm):
             # (ALSAlgorithm.scala:85); here it is crash-resume
             checkpoint_every=int(self.params.get("checkpointEvery", 0)),
             checkpoint_dir=self.params.get("checkpointDir"))
-        device = torch.device("cuda") if torch.cuda.is_available() \
-            else torch.device("cpu")
+        device = pdist.compute_device()
         X, Y = train_als(users, items, vals, len(user_map), len(item_map),
                          p, device=device)
         return ALSModel(p.rank, X, Y, user_map, item_map)

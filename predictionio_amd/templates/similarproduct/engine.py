@@ -29,6 +29,8 @@ from typing import Any, Dict, List, Optional, Set, Tuple
 
 import torch
 
+from predictionio_amd.parallel import dist as pdist
+
 from predictionio_amd.controller import (
     Algorithm, DataSource as BaseDataSource, Engine, EngineFactory,
     Preparator as BasePreparator, SanityCheck, Serving as BaseServing,
@@ -177,8 +179,7 @@ class ALSAlgorithm(Algorithm):
             lambda_=float(self.params.get("lambda", 0.01)),
             alpha=float(self.params.get("alpha", 1.0)),
             implicit=True, seed=self.params.get("seed"))
-        device = torch.device("cuda") if torch.cuda.is_available() \
-            else torch.device("cpu")
+        device = pdist.compute_device()
         _, Y = train_als(users, items, vals, len(user_map), len(item_map),
                          p, device=device)
         Yn = torch.nn.functional.normalize(Y, dim=1, eps=1e-9)
