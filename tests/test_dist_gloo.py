@@ -337,3 +337,23 @@ class TestCheckpointAgreement:
         monkeypatch.setenv("PIO_TEST_CKPT_DIR", str(tmp_path))
         res = _spawn("_ckpt_disagreement", port=29641)
         assert res[0] is True and res[1] is True
+
+
+class TestComputeDevice:
+    def test_cpu_when_no_cuda(self):
+        from predictionio_amd.parallel import dist as pdist
+        assert pdist.compute_device().type == "cpu"
+
+    def test_follows_backend_not_availability(self, monkeypatch):
+        """With a GPU visible but a gloo process group (ranks exceeded
+        the GPU count), compute must stay on CPU — gloo cannot gather
+        CUDA tensors."""
+        import torch
+
+        from predictionio_amd.parallel import dist as pdist
+        monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+        monkeypatch.setattr(pdist, "is_distributed", lambda: True)
+        monkeypatch.setattr(pdist.dist, "get_backend", lambda: "gloo")
+        assert pdist.compute_device().type == "cpu"
+        monkeypatch.setattr(pdist.dist, "get_backend", lambda: "nccl")
+        assert pdist.compute_device().type == "cuda"
