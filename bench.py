@@ -198,6 +198,10 @@ def main():
     if world != args.gpus and "WORLD_SIZE" in os.environ:
         args.gpus = world
     use_cuda = torch.cuda.is_available() and not args.cpu_small
+    if use_cuda and world > 1:
+        # more ranks than GPUs: init_from_env fell back to gloo and the
+        # compute must follow the collectives to the CPU
+        use_cuda = pdist.compute_device().type == "cuda"
     if use_cuda:
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
         torch.cuda.set_device(local_rank)
