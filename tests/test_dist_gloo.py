@@ -340,8 +340,11 @@ class TestCheckpointAgreement:
 
 
 class TestComputeDevice:
-    def test_cpu_when_no_cuda(self):
+    def test_cpu_when_no_cuda(self, monkeypatch):
+        import torch
+
         from predictionio_amd.parallel import dist as pdist
+        monkeypatch.setattr(torch.cuda, "is_available", lambda: False)
         assert pdist.compute_device().type == "cpu"
 
     def test_follows_backend_not_availability(self, monkeypatch):
