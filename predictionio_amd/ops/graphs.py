@@ -86,7 +86,11 @@ class GraphedTopK:
     def close(self) -> None:
         """Tear the graph down SAFELY: destroying a replayed graph with
         in-flight work and then capturing a new one memory-faulted on
-        ROCm (round-2 repro) — synchronize, reset, release."""
+        ROCm (round-2 repro) — synchronize, reset, release. Also drops
+        this graph's pool-allocated buffers: a later capture reusing
+        the shared pool hits an allocator use_count assert if the old
+        blocks are still referenced (callers must likewise not hold
+        output views across close — clone to retain)."""
         gph = getattr(self, "_graph", None)
         if gph is not None:
             try:
@@ -95,6 +99,16 @@ class GraphedTopK:
             except Exception:
                 pass
             self._graph = None
+        for a in ("_out_v", "_out_i", "_xq"):
+            if hasattr(self, a):
+                delattr(self, a)
+        # ROCm's allocator keeps the pool entry referenced even after
+        # reset (use_count assert on the NEXT capture into the same
+        # pool — create->close->create repro); hand the next generation
+        # of graphs a fresh pool instead. Create concurrent graphs
+        # together BEFORE closing any of that generation.
+        global _shared_pool
+        _shared_pool = None
 
     def __del__(self):  # noqa: D105
         try:
