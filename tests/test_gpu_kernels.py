@@ -376,6 +376,29 @@ class TestTemplatesOnGPU:
 
 @requires_gpu
 class TestGraphedTopK:
+    def test_create_close_create(self):
+        """Graph lifecycle: close() must fully release the capture so a
+        LATER graph can be built (ROCm kept the pool entry referenced
+        after reset — allocator use_count assert on the next capture
+        into the same pool; close() now retires the shared pool)."""
+        from predictionio_amd.ops.graphs import GraphedTopK
+        Y = torch.randn((50_000, 64)).float().cuda()
+        for _ in range(2):
+            gt = GraphedTopK(Y, 10, 4)
+            v, i = gt(torch.randn((4, 64), device="cuda"))
+            assert i.shape == (4, 10)
+            del v, i
+            gt.close()
+        # concurrent generation, closed together, then a fresh one
+        a, b = GraphedTopK(Y, 10, 1), GraphedTopK(Y, 10, 8)
+        a(torch.randn((1, 64), device="cuda"))
+        b(torch.randn((8, 64), device="cuda"))
+        a.close()
+        b.close()
+        c = GraphedTopK(Y, 10, 2)
+        c(torch.randn((2, 64), device="cuda"))
+        c.close()
+
     def test_graph_matches_eager(self):
         """hipGraph-captured serving step equals the eager launch and is
         replayable with new query content."""
