@@ -30,6 +30,28 @@ def _load_variant(engine_dir: str, variant: str) -> dict:
         return json.load(f)
 
 
+def _check_template_min_version(engine_dir: str) -> None:
+    """template.json minimum-version gate, honored by build AND train
+    like the reference (Template.verifyTemplateMinVersion called from
+    commands/Engine.scala:148-151 and :188-190)."""
+    tj = os.path.join(engine_dir, "template.json")
+    if not os.path.exists(tj):
+        return
+    import predictionio_amd
+    with open(tj) as f:
+        meta = json.load(f)
+    need = (meta.get("pio", {}).get("version", {}) or {}).get("min")
+    if need:
+        have = tuple(int(x) for x in
+                     predictionio_amd.__version__.split("."))
+        want = tuple(int(x) for x in str(need).split("."))
+        if have < want:
+            click.echo(f"[ERROR] This template requires PIO >= "
+                       f"{need}; installed "
+                       f"{predictionio_amd.__version__}.")
+            raise SystemExit(1)
+
+
 def _add_engine_dir(engine_dir: str) -> None:
     d = os.path.abspath(engine_dir)
     if d not in sys.path:
@@ -89,21 +111,7 @@ def build(engine_dir):
     Honors the template.json minimum-version gate
     (Template.verifyTemplateMinVersion, commands/Template.scala:58)."""
     _add_engine_dir(engine_dir)
-    tj = os.path.join(engine_dir, "template.json")
-    if os.path.exists(tj):
-        import predictionio_amd
-        with open(tj) as f:
-            meta = json.load(f)
-        need = (meta.get("pio", {}).get("version", {}) or {}).get("min")
-        if need:
-            have = tuple(int(x) for x in
-                         predictionio_amd.__version__.split("."))
-            want = tuple(int(x) for x in str(need).split("."))
-            if have < want:
-                click.echo(f"[ERROR] This template requires PIO >= "
-                           f"{need}; installed "
-                           f"{predictionio_amd.__version__}.")
-                raise SystemExit(1)
+    _check_template_min_version(engine_dir)
     from predictionio_amd.ops import build as ops_build
     click.echo("[INFO] building HIP extension (gfx950)...")
     ops_build.build()
@@ -119,7 +127,10 @@ def build(engine_dir):
 @click.option("--skip-sanity-check", is_flag=True)
 @click.option("--gpus", default=0, help="train with N GPUs via torchrun")
 def train(engine_dir, variant, batch, skip_sanity_check, gpus):
-    """Train an engine instance (RunWorkflow → CreateWorkflow.main)."""
+    """Train an engine instance (RunWorkflow → CreateWorkflow.main).
+    Honors the template.json minimum-version gate like the reference's
+    train path (commands/Engine.scala:188-190)."""
+    _check_template_min_version(engine_dir)
     if gpus > 1:
         import socket
         import subprocess

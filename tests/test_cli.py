@@ -151,3 +151,17 @@ class TestTemplateMinVersion:
         r = CliRunner().invoke(cli, ["build", "--engine-dir", str(d)])
         assert r.exit_code == 1
         assert "requires PIO" in r.output
+
+    def test_train_rejects_newer_min_version(self, tmp_path):
+        """The reference checks the gate on train too
+        (commands/Engine.scala:188-190)."""
+        import json
+        from click.testing import CliRunner
+        from predictionio_amd.cli.main import cli
+        d = tmp_path / "eng"
+        d.mkdir()
+        (d / "template.json").write_text(
+            json.dumps({"pio": {"version": {"min": "99.0.0"}}}))
+        r = CliRunner().invoke(cli, ["train", "--engine-dir", str(d)])
+        assert r.exit_code == 1
+        assert "requires PIO" in r.output
