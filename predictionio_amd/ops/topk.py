@@ -153,12 +153,16 @@ def _topk_score_mfma(Xq: torch.Tensor, Y: torch.Tensor, K: int,
     pf = _mfma_rank(f)
     Xp = Xq if pf == f else torch.nn.functional.pad(Xq, (0, pf - f))
     if n_slices is None:
-        # insert volume grows ~linearly with n_slices (each slice resets
-        # the per-query threshold), so use just enough slices to fill
-        # the chip (measured sweep: scripts/mfma_phase_probe.py)
+        # just enough slices to fill the chip ONCE: the resident-WG
+        # count depends on the kernel's LDS (64-item tile + the
+        # per-query top-K lists), so the target shrinks for large K —
+        # a 1536-WG grid at K=64 only fits 768 at a time and the
+        # second pass doubled the time (profiles/serve_k_sweep_r2.log)
         upw = 128 if os.environ.get("PIO_TOPK_QB") == "2" else 64
         ublocks = (B + upw - 1) // upw
-        n_slices = max(2, min(1536 // ublocks, (N + 255) // 256))
+        lds = 64 * 2 * pf + 8 * (upw // 64) * 64 * (K + 1) + 4 * 64
+        target = 256 * max(1, min(6, (160 * 1024) // lds))
+        n_slices = max(2, min(target // ublocks, (N + 255) // 256))
     # Y's bf16 copy is cached (factors are static across serving batches);
     # Xq is cast per call — it changes every batch, and under hipGraph
     # capture (GraphedTopK) the cast must be part of the captured work
